@@ -1,0 +1,96 @@
+"""Actor — CPU rollout loop.
+
+Owns local fp32 CPU copies of ActorNet + CriticNet (the critic provides
+step-wise Q values for initial priorities), steps the env with
+a = mu(o, h) + N(0, sigma_i), maintains both LSTM states across steps,
+feeds the sequence cutter, and periodically pulls fresh learner weights.
+
+Runs in-process (single-process mode, tests) or as the body of an actor
+process (parallel/actor_pool.py).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import numpy as np
+import torch
+
+from .config import Config
+from .envs import make_env
+from .models import make_models
+from .replay.sequence import Sequence, SequenceCutter
+from .utils.noise import GaussianNoise, sigma_ladder
+
+
+class ActorWorker:
+    def __init__(self, cfg: Config, actor_id: int = 0, sigma: Optional[float] = None,
+                 eval_mode: bool = False):
+        self.cfg = cfg
+        self.actor_id = actor_id
+        self.eval_mode = eval_mode
+        self.env = make_env(cfg.env.name, seed=cfg.env.seed + 1000 * actor_id + 1,
+                            time_limit=cfg.env.time_limit)
+        spec = self.env.spec
+        self.spec = spec
+        self.actor_net, self.critic_net, _, _ = make_models(spec, cfg.model, device="cpu")
+        self.actor_net.eval()
+        self.critic_net.eval()
+        if sigma is None:
+            sigma = 0.0 if eval_mode else sigma_ladder(
+                actor_id, max(cfg.actor.num_actors, 1), cfg.actor.sigma_max, cfg.actor.sigma_alpha)
+        self.sigma = sigma
+        self.noise = GaussianNoise(spec.action_dim, sigma, seed=cfg.env.seed + actor_id + 7777)
+        R = cfg.replay
+        self.cutter = SequenceCutter(R.seq_len, R.overlap, R.burn_in, R.n_step,
+                                     R.gamma, R.priority_eta, cfg.model.hidden_size)
+        self._obs = self.env.reset()
+        H = cfg.model.hidden_size
+        self._state = (torch.zeros(1, H), torch.zeros(1, H),
+                       torch.zeros(1, H), torch.zeros(1, H))  # h_a, c_a, h_c, c_c
+        self.episode_return = 0.0
+        self.episode_returns: List[float] = []
+        self.env_steps = 0
+
+    # -------------------------------------------------------------- weights
+    def load_weights(self, actor_sd: dict, critic_sd: dict) -> None:
+        self.actor_net.load_state_dict(actor_sd)
+        self.critic_net.load_state_dict(critic_sd)
+
+    # ----------------------------------------------------------------- step
+    @torch.no_grad()
+    def rollout(self, num_steps: int) -> List[Sequence]:
+        """Run `num_steps` env steps; return the sequences cut along the way."""
+        out: List[Sequence] = []
+        h_a, c_a, h_c, c_c = self._state
+        for _ in range(num_steps):
+            snapshot = np.stack([h_a[0].numpy(), c_a[0].numpy(),
+                                 h_c[0].numpy(), c_c[0].numpy()])
+            obs_t = torch.from_numpy(np.ascontiguousarray(self._obs)).unsqueeze(0)
+            if self.spec.pixel:
+                a_feat = obs_t
+            else:
+                a_feat = obs_t.float()
+            mu, (h_a, c_a) = self.actor_net.step(a_feat, (h_a, c_a))
+            a = mu[0].numpy()
+            if not self.eval_mode:
+                a = np.clip(a + self.noise(),
+                            -self.spec.action_scale, self.spec.action_scale)
+            q, (h_c, c_c) = self.critic_net.step(
+                a_feat, torch.from_numpy(a).unsqueeze(0), (h_c, c_c))
+            next_obs, r, done, trunc = self.env.step(a)
+            self.episode_return += r
+            self.env_steps += 1
+            out.extend(self.cutter.push(self._obs, a, r, done, float(q[0]), snapshot))
+            self._obs = next_obs
+            if done or trunc:
+                if trunc and not done:
+                    out.extend(self.cutter.flush())
+                self.episode_returns.append(self.episode_return)
+                self.episode_return = 0.0
+                self._obs = self.env.reset()
+                H = self.cfg.model.hidden_size
+                h_a = torch.zeros(1, H); c_a = torch.zeros(1, H)
+                h_c = torch.zeros(1, H); c_c = torch.zeros(1, H)
+        self._state = (h_a, c_a, h_c, c_c)
+        return out

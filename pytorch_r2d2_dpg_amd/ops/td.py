@@ -1,0 +1,46 @@
+"""n-step TD-target reduction + sequence priorities (kernel K3).
+
+One fused, memory-bound pass on GPU: for each sequence b
+and training-window step t,
+
+    folded_t = sum_{k<n} gamma^k r_{t+k}   (truncated at the first done)
+    mask_t   = 0 if a done occurs in [t, t+n)
+    y_t      = folded_t + gamma^n * mask_t * q'_{t+n}
+    delta_t  = y_t - q_t
+    p_b      = eta * max_t|delta_t| + (1-eta) * mean_t|delta_t|
+
+Inputs carry the n-step tail: rewards/dones are [B, T+n-ish] (>= T+n-1 used),
+bootstrap q' is [B, T] (already the value at t+n from the target-net unroll).
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from . import hip_ext
+from .reference import fold_nstep_ref, sequence_priority_ref
+
+
+def td_targets_and_priorities(
+    rewards: torch.Tensor,        # [B, T_ext] raw rewards from window start
+    dones: torch.Tensor,          # [B, T_ext] terminal flags (1.0 = terminated)
+    bootstrap_q: torch.Tensor,    # [B, T] Q'(s_{t+n}, mu'(s_{t+n}))
+    q_online: torch.Tensor,       # [B, T] Q(s_t, a_t), detached
+    gamma: float,
+    n_step: int,
+    eta: float,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Returns (targets y [B,T], |delta| [B,T], priorities [B])."""
+    T = q_online.shape[1]
+    if rewards.is_cuda:
+        return hip_ext().td_targets(
+            rewards.contiguous(), dones.contiguous(), bootstrap_q.contiguous(),
+            q_online.contiguous(), float(gamma), int(n_step), float(eta),
+        )
+    folded, mask = fold_nstep_ref(rewards, dones, gamma, n_step, T)
+    y = folded + (gamma ** n_step) * mask * bootstrap_q
+    td_abs = (y - q_online).abs()
+    prio = sequence_priority_ref(td_abs, eta)
+    return y, td_abs, prio

@@ -1,0 +1,128 @@
+"""Actor-side sequence cutter + local-priority computation.
+
+Maintains the rolling transition window and the recurrent-state snapshots;
+every ``stride = seq_len - overlap`` steps emits a fixed-length sequence of
+``stored_len = seq_len + n_step`` transitions (the n-step tail rides along so
+the learner's fused K3 kernel can fold raw rewards on GPU), carrying the
+(actor, critic) LSTM states snapshotted at the sequence's first step and an
+initial priority from the actor's own step-wise Q values:
+
+    delta_t = fold_n(r)_t + gamma^n * mask_t * q_{t+n} - q_t
+    p0      = eta * max|delta| + (1-eta) * mean|delta|
+
+At episode end the window is flushed (right-padded to fixed length with
+zero-reward terminal steps, mask excluded via done flags) and reset.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import numpy as np
+
+
+@dataclass
+class Sequence:
+    obs: np.ndarray        # [L, *obs_shape]
+    action: np.ndarray     # [L, A]
+    reward: np.ndarray     # [L]
+    done: np.ndarray       # [L] 1.0 where episode terminated at that step
+    state: np.ndarray      # [4, H]: h_actor, c_actor, h_critic, c_critic at t=0
+    priority: float
+    length: int            # number of real (non-pad) steps
+
+
+class SequenceCutter:
+    def __init__(self, seq_len: int, overlap: int, burn_in: int, n_step: int,
+                 gamma: float, eta: float, hidden_size: int):
+        assert 0 <= overlap < seq_len
+        self.seq_len = seq_len
+        self.stride = seq_len - overlap
+        self.burn_in = burn_in
+        self.n_step = n_step
+        self.stored_len = seq_len + n_step
+        self.gamma = gamma
+        self.eta = eta
+        self.hidden_size = hidden_size
+        self._reset_window()
+
+    def _reset_window(self):
+        self.obs: List[np.ndarray] = []
+        self.act: List[np.ndarray] = []
+        self.rew: List[float] = []
+        self.done: List[float] = []
+        self.q: List[float] = []
+        self.states: List[np.ndarray] = []  # [4, H] snapshot BEFORE each step
+
+    def push(self, obs, action, reward, done, q_value, state_snapshot) -> List[Sequence]:
+        """Add one transition; returns 0..1 emitted sequences."""
+        self.obs.append(np.asarray(obs))
+        self.act.append(np.asarray(action, dtype=np.float32))
+        self.rew.append(float(reward))
+        self.done.append(1.0 if done else 0.0)
+        self.q.append(float(q_value))
+        self.states.append(np.asarray(state_snapshot, dtype=np.float32))
+        out: List[Sequence] = []
+        if len(self.obs) >= self.stored_len:
+            out.append(self._emit(self.stored_len))
+            # slide by stride: keep the overlap + tail for the next sequence
+            k = self.stride
+            self.obs = self.obs[k:]
+            self.act = self.act[k:]
+            self.rew = self.rew[k:]
+            self.done = self.done[k:]
+            self.q = self.q[k:]
+            self.states = self.states[k:]
+        if done:
+            seqs = self.flush()
+            out.extend(seqs)
+        return out
+
+    def flush(self) -> List[Sequence]:
+        """Emit the remaining window (padded) at episode end, then reset."""
+        out: List[Sequence] = []
+        # only emit if there is at least one step beyond burn-in to train on
+        if len(self.obs) > self.burn_in + 1:
+            out.append(self._emit(len(self.obs)))
+        self._reset_window()
+        return out
+
+    def _emit(self, length: int) -> Sequence:
+        L = self.stored_len
+        obs_arr = np.zeros((L,) + self.obs[0].shape, dtype=self.obs[0].dtype)
+        act_arr = np.zeros((L, self.act[0].shape[0]), dtype=np.float32)
+        rew_arr = np.zeros(L, dtype=np.float32)
+        done_arr = np.ones(L, dtype=np.float32)  # pad region counts as done
+        n = min(length, L)
+        obs_arr[:n] = np.stack(self.obs[:n])
+        act_arr[:n] = np.stack(self.act[:n])
+        rew_arr[:n] = np.asarray(self.rew[:n], dtype=np.float32)
+        done_arr[:n] = np.asarray(self.done[:n], dtype=np.float32)
+        if n < L:
+            obs_arr[n:] = obs_arr[n - 1]  # repeat last obs into the pad
+        prio = self._initial_priority(rew_arr, done_arr, np.asarray(self.q[:n], dtype=np.float32), n)
+        return Sequence(obs_arr, act_arr, rew_arr, done_arr,
+                        self.states[0].copy(), float(prio), n)
+
+    def _initial_priority(self, rew, done, q, n_real) -> float:
+        """n-step TD error against the actor's own step-wise Q estimates."""
+        t_lo = self.burn_in
+        t_hi = max(t_lo + 1, min(self.seq_len, n_real))
+        deltas = []
+        for t in range(t_lo, t_hi):
+            acc, g, alive = 0.0, 1.0, 1.0
+            for k in range(self.n_step):
+                if t + k >= n_real:
+                    alive = 0.0
+                    break
+                acc += g * rew[t + k]
+                alive *= 1.0 - done[t + k]
+                g *= self.gamma
+            boot = q[t + self.n_step] if (t + self.n_step) < n_real else 0.0
+            y = acc + (self.gamma ** self.n_step) * alive * boot
+            deltas.append(abs(y - q[t]) if t < n_real else 0.0)
+        d = np.asarray(deltas, dtype=np.float64)
+        if d.size == 0:
+            return 1.0
+        return self.eta * d.max() + (1.0 - self.eta) * d.mean()

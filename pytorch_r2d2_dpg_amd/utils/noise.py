@@ -1,0 +1,28 @@
+"""Exploration noise.
+
+Ape-X-style per-actor scale ladder: sigma_i = sigma_max^(1 + alpha*i/(N-1)),
+so actors explore at geometrically spaced intensities; each actor adds
+iid Gaussian noise of its sigma to the deterministic action.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+
+def sigma_ladder(actor_id: int, num_actors: int, sigma_max: float = 0.4,
+                 alpha: float = 7.0) -> float:
+    if num_actors <= 1:
+        return sigma_max
+    expo = 1.0 + alpha * actor_id / (num_actors - 1)
+    return float(sigma_max ** expo)
+
+
+class GaussianNoise:
+    def __init__(self, action_dim: int, sigma: float, seed: int = 0):
+        self.sigma = sigma
+        self.action_dim = action_dim
+        self._rng = np.random.default_rng(seed)
+
+    def __call__(self) -> np.ndarray:
+        return self._rng.standard_normal(self.action_dim).astype(np.float32) * self.sigma

@@ -156,9 +156,9 @@ class Learner:
             self.polyak_actor(cfg.learner.tau)
             self.polyak_critic(cfg.learner.tau)
 
-        # ---- priority write-back
-        if self.replay is not None:
-            self.replay.update_priorities(batch["idx"], prio.cpu().numpy())
+        # ---- priority write-back (stays on-GPU with the K6 tree kernel)
+        if self.replay is not None and batch.get("idx") is not None:
+            self.replay.update_priorities(batch["idx"], prio)
 
         self.step_count += 1
         return StepStats(

@@ -1,0 +1,236 @@
+// Python bindings for the MI355X HIP kernels (K1-K7).
+// Pure C++ TU: talks to the .hip translation units through extern "C"
+// launchers taking raw pointers + the current HIP stream from torch.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <hip/hip_runtime.h>
+
+#include <tuple>
+
+// ---- extern "C" launchers from the .hip TUs ------------------------------
+extern "C" int launch_lstm_fwd_bf16(const void*, const void*, const void*, const void*,
+                                    void*, void*, void*, float*, int, int, int, hipStream_t);
+extern "C" int launch_lstm_fwd_f32(const void*, const void*, const void*, const void*,
+                                   void*, void*, void*, float*, int, int, int, hipStream_t);
+extern "C" int launch_lstm_bwd_bf16(const void*, const void*, const void*, const void*,
+                                    const float*, const void*, const void*, void*, void*,
+                                    void*, int, int, int, hipStream_t);
+extern "C" int launch_lstm_bwd_f32(const void*, const void*, const void*, const void*,
+                                   const float*, const void*, const void*, void*, void*,
+                                   void*, int, int, int, hipStream_t);
+extern "C" void launch_td_targets(const float*, const float*, const float*, const float*,
+                                  float*, float*, float*, int, int, int, int, float, float,
+                                  int, float, hipStream_t);
+extern "C" void launch_polyak_multi(const unsigned long long*, const long*, int, long,
+                                    float, int, hipStream_t);
+extern "C" void launch_sumtree_sample(const float*, int, int, unsigned long long,
+                                      unsigned long long, long*, float*, hipStream_t);
+extern "C" void launch_sumtree_update(float*, int, const long*, const float*, int,
+                                      hipStream_t);
+extern "C" void launch_fp8_encode(const float*, unsigned char*, long, float, hipStream_t);
+extern "C" void launch_fp8_gather_decode(const unsigned char*, const long*, void*, int,
+                                         long, int, float, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_dev_contig(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ---- K1: fused LSTM sequence forward -------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
+             torch::Tensor wh) {
+    check_dev_contig(gates_x, "gates_x");
+    check_dev_contig(h0, "h0");
+    check_dev_contig(c0, "c0");
+    check_dev_contig(wh, "wh");
+    const auto T = gates_x.size(0);
+    const auto B = gates_x.size(1);
+    const auto H = h0.size(1);
+    TORCH_CHECK(gates_x.size(2) == 4 * H, "gates_x last dim must be 4H");
+    TORCH_CHECK(H % 64 == 0, "hidden size must be a multiple of 64 (got ", H, ")");
+    auto dt = gates_x.scalar_type();
+    TORCH_CHECK(dt == torch::kBFloat16 || dt == torch::kFloat32,
+                "lstm_seq_fwd supports bf16/fp32 only");
+    TORCH_CHECK(h0.scalar_type() == dt && wh.scalar_type() == dt && c0.scalar_type() == dt,
+                "dtype mismatch");
+
+    auto h_seq = torch::empty({T, B, H}, gates_x.options());
+    auto c_final = torch::empty({B, H}, gates_x.options());
+    auto gates_act = torch::empty({T, B, 4 * H}, gates_x.options());
+    auto c_seq = torch::empty({T, B, H}, gates_x.options().dtype(torch::kFloat32));
+
+    int rc;
+    if (dt == torch::kBFloat16)
+        rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
+                                  wh.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
+                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
+                                  (int)T, (int)B, (int)H, cur_stream());
+    else
+        rc = launch_lstm_fwd_f32(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
+                                 wh.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
+                                 gates_act.data_ptr(), c_seq.data_ptr<float>(),
+                                 (int)T, (int)B, (int)H, cur_stream());
+    TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);
+    return {h_seq, c_final, gates_act, c_seq};
+}
+
+// ---- K2: fused LSTM sequence backward -------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
+lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
+             torch::Tensor gates_act, torch::Tensor c_seq, torch::Tensor c0,
+             torch::Tensor whT) {
+    check_dev_contig(dh_seq, "dh_seq");
+    check_dev_contig(dhT, "dhT");
+    check_dev_contig(dcT, "dcT");
+    check_dev_contig(gates_act, "gates_act");
+    check_dev_contig(c_seq, "c_seq");
+    check_dev_contig(c0, "c0");
+    check_dev_contig(whT, "whT");
+    const auto T = dh_seq.size(0);
+    const auto B = dh_seq.size(1);
+    const auto H = dh_seq.size(2);
+    TORCH_CHECK(whT.size(0) == H && whT.size(1) == 4 * H, "whT must be [H, 4H]");
+    auto dt = dh_seq.scalar_type();
+
+    auto dgates = torch::empty({T, B, 4 * H}, dh_seq.options());
+    auto dh0 = torch::empty({B, H}, dh_seq.options());
+    auto dc0 = torch::empty({B, H}, dh_seq.options());
+
+    int rc;
+    if (dt == torch::kBFloat16)
+        rc = launch_lstm_bwd_bf16(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
+                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
+                                  c0.data_ptr(), whT.data_ptr(), dgates.data_ptr(),
+                                  dh0.data_ptr(), dc0.data_ptr(),
+                                  (int)T, (int)B, (int)H, cur_stream());
+    else
+        rc = launch_lstm_bwd_f32(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
+                                 gates_act.data_ptr(), c_seq.data_ptr<float>(),
+                                 c0.data_ptr(), whT.data_ptr(), dgates.data_ptr(),
+                                 dh0.data_ptr(), dc0.data_ptr(),
+                                 (int)T, (int)B, (int)H, cur_stream());
+    TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);
+    return {dgates, dh0, dc0};
+}
+
+// ---- K3: fused n-step TD targets + priorities -----------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
+td_targets(torch::Tensor rewards, torch::Tensor dones, torch::Tensor q_boot,
+           torch::Tensor q_online, double gamma, int64_t n_step, double eta,
+           bool rescale = false, double rescale_eps = 1e-3) {
+    check_dev_contig(rewards, "rewards");
+    check_dev_contig(dones, "dones");
+    check_dev_contig(q_boot, "q_boot");
+    check_dev_contig(q_online, "q_online");
+    const auto B = q_online.size(0);
+    const auto T = q_online.size(1);
+    const auto t_ext = rewards.size(1);
+    TORCH_CHECK(rewards.scalar_type() == torch::kFloat32, "rewards must be fp32");
+    auto y = torch::empty({B, T}, q_online.options());
+    auto td_abs = torch::empty({B, T}, q_online.options());
+    auto prio = torch::empty({B}, q_online.options());
+    launch_td_targets(rewards.data_ptr<float>(), dones.data_ptr<float>(),
+                      q_boot.data_ptr<float>(), q_online.data_ptr<float>(),
+                      y.data_ptr<float>(), td_abs.data_ptr<float>(),
+                      prio.data_ptr<float>(), (int)B, (int)T, (int)t_ext,
+                      (int)n_step, (float)gamma, (float)eta,
+                      rescale ? 1 : 0, (float)rescale_eps, cur_stream());
+    return {y, td_abs, prio};
+}
+
+// ---- K4: fused multi-tensor Polyak ----------------------------------------
+void polyak_multi(torch::Tensor ptr_table, torch::Tensor off_table, int64_t total,
+                  double tau, bool is_bf16) {
+    check_dev_contig(ptr_table, "ptr_table");
+    check_dev_contig(off_table, "off_table");
+    int n = (int)(off_table.size(0) - 1);
+    launch_polyak_multi(
+        reinterpret_cast<const unsigned long long*>(ptr_table.data_ptr<int64_t>()),
+        reinterpret_cast<const long*>(off_table.data_ptr<int64_t>()), n,
+        (long)total, (float)tau, is_bf16 ? 1 : 0, cur_stream());
+}
+
+// ---- K5: sum-tree stratified sample ---------------------------------------
+std::tuple<torch::Tensor, torch::Tensor>
+sumtree_sample(torch::Tensor tree, int64_t cap, int64_t batch, int64_t seed,
+               int64_t step) {
+    check_dev_contig(tree, "tree");
+    TORCH_CHECK(tree.scalar_type() == torch::kFloat32, "tree must be fp32");
+    auto idx = torch::empty({batch}, tree.options().dtype(torch::kInt64));
+    auto prob = torch::empty({batch}, tree.options());
+    launch_sumtree_sample(tree.data_ptr<float>(), (int)cap, (int)batch,
+                          (unsigned long long)seed, (unsigned long long)step,
+                          (long*)idx.data_ptr<int64_t>(), prob.data_ptr<float>(),
+                          cur_stream());
+    return {idx, prob};
+}
+
+// ---- K6: sum-tree leaf update + level-synchronous propagate ---------------
+void sumtree_update(torch::Tensor tree, int64_t cap, torch::Tensor idx,
+                    torch::Tensor prio) {
+    check_dev_contig(tree, "tree");
+    check_dev_contig(idx, "idx");
+    check_dev_contig(prio, "prio");
+    TORCH_CHECK(idx.scalar_type() == torch::kInt64 && prio.scalar_type() == torch::kFloat32,
+                "idx must be int64, prio fp32");
+    launch_sumtree_update(tree.data_ptr<float>(), (int)cap,
+                          (const long*)idx.data_ptr<int64_t>(),
+                          prio.data_ptr<float>(), (int)idx.size(0), cur_stream());
+}
+
+// ---- K7: fp8 e4m3 encode / fused gather+decode ----------------------------
+void fp8_encode(torch::Tensor src, torch::Tensor dst, double scale) {
+    check_dev_contig(src, "src");
+    check_dev_contig(dst, "dst");
+    TORCH_CHECK(src.scalar_type() == torch::kFloat32 && dst.scalar_type() == torch::kUInt8,
+                "fp8_encode: src fp32 -> dst u8");
+    TORCH_CHECK(src.numel() == dst.numel(), "numel mismatch");
+    launch_fp8_encode(src.data_ptr<float>(), dst.data_ptr<uint8_t>(), src.numel(),
+                      (float)scale, cur_stream());
+}
+
+torch::Tensor fp8_gather_decode(torch::Tensor store, torch::Tensor idx,
+                                torch::ScalarType out_dtype, double inv_scale) {
+    check_dev_contig(store, "store");
+    check_dev_contig(idx, "idx");
+    TORCH_CHECK(store.scalar_type() == torch::kUInt8, "store must be u8 (fp8 e4m3)");
+    const auto B = idx.size(0);
+    long row = 1;
+    for (int d = 1; d < store.dim(); ++d) row *= store.size(d);
+    std::vector<int64_t> shape;
+    shape.push_back(B);
+    for (int d = 1; d < store.dim(); ++d) shape.push_back(store.size(d));
+    auto out = torch::empty(shape, store.options().dtype(out_dtype));
+    launch_fp8_gather_decode(store.data_ptr<uint8_t>(),
+                             (const long*)idx.data_ptr<int64_t>(), out.data_ptr(),
+                             out_dtype == torch::kBFloat16 ? 1 : 0, row, (int)B,
+                             (float)inv_scale, cur_stream());
+    return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("lstm_seq_fwd", &lstm_seq_fwd, "fused LSTM sequence forward (K1)");
+    m.def("lstm_seq_bwd", &lstm_seq_bwd, "fused LSTM sequence backward (K2)");
+    m.def("td_targets", &td_targets, "fused n-step TD targets + priorities (K3)",
+          py::arg("rewards"), py::arg("dones"), py::arg("q_boot"), py::arg("q_online"),
+          py::arg("gamma"), py::arg("n_step"), py::arg("eta"),
+          py::arg("rescale") = false, py::arg("rescale_eps") = 1e-3);
+    m.def("polyak_multi", &polyak_multi, "fused multi-tensor Polyak update (K4)");
+    m.def("sumtree_sample", &sumtree_sample, "sum-tree stratified sample (K5)");
+    m.def("sumtree_update", &sumtree_update, "sum-tree leaf update + propagate (K6)");
+    m.def("fp8_encode", &fp8_encode, "fp8 e4m3 encode (K7)");
+    m.def("fp8_gather_decode", &fp8_gather_decode, "fp8 fused gather+decode (K7)",
+          py::arg("store"), py::arg("idx"), py::arg("out_dtype"),
+          py::arg("inv_scale") = 1.0);
+}

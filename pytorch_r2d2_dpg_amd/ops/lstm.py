@@ -3,10 +3,10 @@
 MI355X-native split of the work:
 
 * input projection ``gates_x = x @ Wx^T + b`` — one large, fully-parallel
-  library GEMM over [T*B, I] (rocBLAS/hipBLASLt via torch.matmul),
+  library GEMM over [T*B, I] (rocBLAS/hipBLASLt via torch.addmm),
 * the sequential recurrence — ONE persistent HIP kernel launch per direction
   (``lstm_seq_fwd`` / ``lstm_seq_bwd``): each workgroup owns a tile of batch
-  rows, keeps its (h, c) tile in LDS across all T timesteps, streams Wh
+  rows, keeps its (h, c) tile on-CU across all T timesteps, streams Wh
   through the XCD's L2, and computes the per-step ``h @ Wh^T`` on MFMA.
   This replaces the reference's per-timestep ``nn.LSTM`` launches (the
   dominant launch-bound cost — SURVEY.md §3.3),
@@ -18,7 +18,7 @@ directly; on GPU the HIP extension is mandatory (ops/__init__.py).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 
@@ -27,7 +27,11 @@ from .reference import lstm_seq_ref
 
 
 class _FusedLSTMSeq(torch.autograd.Function):
-    """GPU path. Saves post-activation gates + cell states for backward."""
+    """GPU path. Saves post-activation gates + cell states for backward.
+
+    Outputs (h_seq, c_final) only; the caller takes hT = h_seq[-1] so its
+    gradient folds into dh_seq through the select op (no view-output issues).
+    """
 
     @staticmethod
     def forward(ctx, x, h0, c0, wx, wh, b):
@@ -36,20 +40,23 @@ class _FusedLSTMSeq(torch.autograd.Function):
         ext = hip_ext()
         # library GEMM: [T*B, I] @ [I, 4H] (+bias)
         gates_x = torch.addmm(b, x.reshape(T * B, I), wx.t()).view(T, B, 4 * H)
-        h_seq, c_final, gates_act, c_seq = ext.lstm_seq_fwd(gates_x, h0, c0, wh)
+        h_seq, c_final, gates_act, c_seq = ext.lstm_seq_fwd(
+            gates_x.contiguous(), h0.contiguous(), c0.contiguous(), wh.contiguous()
+        )
         ctx.save_for_backward(x, h0, c0, wx, wh, h_seq, gates_act, c_seq)
-        return h_seq, h_seq[-1], c_final
+        return h_seq, c_final
 
     @staticmethod
-    def backward(ctx, dh_seq, dhT, dcT):
+    def backward(ctx, dh_seq, dcT):
         x, h0, c0, wx, wh, h_seq, gates_act, c_seq = ctx.saved_tensors
         T, B, I = x.shape
         H = h0.shape[1]
         ext = hip_ext()
-        dh_seq = dh_seq.contiguous()
-        # persistent reverse-time kernel: dgates [T,B,4H], dh0, dc0
+        zero_hT = torch.zeros_like(h0)
+        whT = wh.t().contiguous()
         dgates, dh0, dc0 = ext.lstm_seq_bwd(
-            dh_seq, dhT.contiguous(), dcT.contiguous(), gates_act, c_seq, c0, wh
+            dh_seq.contiguous(), zero_hT, dcT.contiguous(), gates_act, c_seq,
+            c0.contiguous(), whT,
         )
         dg_flat = dgates.reshape(T * B, 4 * H)
         # library GEMMs for the dense gradients
@@ -57,7 +64,7 @@ class _FusedLSTMSeq(torch.autograd.Function):
         dwx = dg_flat.t() @ x.reshape(T * B, I)
         h_prev = torch.cat([h0.unsqueeze(0), h_seq[:-1]], dim=0).reshape(T * B, H)
         dwh = dg_flat.t() @ h_prev
-        db = dg_flat.sum(dim=0)
+        db = dg_flat.float().sum(dim=0).to(dg_flat.dtype)
         return dx, dh0, dc0, dwx, dwh, db
 
 
@@ -74,5 +81,6 @@ def lstm_seq(
     Returns (h_seq [T,B,H], hT [B,H], cT [B,H]).
     """
     if x.is_cuda:
-        return _FusedLSTMSeq.apply(x, h0, c0, wx, wh, b)
+        h_seq, cT = _FusedLSTMSeq.apply(x, h0, c0, wx, wh, b)
+        return h_seq, h_seq[-1], cT
     return lstm_seq_ref(x, h0, c0, wx, wh, b)

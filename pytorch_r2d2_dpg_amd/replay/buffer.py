@@ -103,6 +103,10 @@ class SequenceReplay:
         }
 
     # ------------------------------------------------------- priority write
-    def update_priorities(self, idx: np.ndarray, priorities: np.ndarray) -> None:
+    def update_priorities(self, idx, priorities) -> None:
+        if torch.is_tensor(idx):
+            idx = idx.cpu().numpy()
+        if torch.is_tensor(priorities):
+            priorities = priorities.detach().cpu().numpy()
         p = np.maximum(np.asarray(priorities, dtype=np.float64), 1e-6)
         self.tree.update(np.asarray(idx, dtype=np.int64), p ** self.per_alpha)

@@ -1,0 +1,278 @@
+"""GPU kernel numerics tests (SURVEY.md §4 "Kernel unit"): every HIP kernel
+vs the plain-PyTorch fp32 reference on identical inputs."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from pytorch_r2d2_dpg_amd.ops.reference import (
+    fold_nstep_ref, lstm_seq_ref, sequence_priority_ref,
+)
+from pytorch_r2d2_dpg_amd.replay.sumtree_cpu import SumTree
+
+
+def ext():
+    from pytorch_r2d2_dpg_amd.ops import hip_ext
+    return hip_ext()
+
+
+def _mk_lstm(T, B, I, H, dtype, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(T, B, I, generator=g)
+    h0 = torch.randn(B, H, generator=g) * 0.5
+    c0 = torch.randn(B, H, generator=g) * 0.5
+    wx = torch.randn(4 * H, I, generator=g) / (I ** 0.5)
+    wh = torch.randn(4 * H, H, generator=g) / (H ** 0.5)
+    b = torch.randn(4 * H, generator=g) * 0.1
+    cpu = (x, h0, c0, wx, wh, b)
+    gpu = tuple(t.cuda().to(dtype) for t in cpu)
+    return cpu, gpu
+
+
+# ------------------------------------------------------------------ K1 fwd
+@pytest.mark.parametrize("H", [64, 128, 256, 512])
+def test_lstm_fwd_f32_vs_ref(H):
+    T, B, I = 7, 24, 33
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.float32)
+    h_ref, hT_ref, cT_ref = lstm_seq_ref(*cpu)
+    x, h0, c0, wx, wh, b = gpu
+    gates_x = torch.addmm(b, x.reshape(T * B, I), wx.t()).view(T, B, 4 * H)
+    h_seq, c_final, gates_act, c_seq = ext().lstm_seq_fwd(gates_x, h0, c0, wh)
+    assert torch.allclose(h_seq.cpu(), h_ref, atol=2e-5), \
+        f"max err {(h_seq.cpu() - h_ref).abs().max()}"
+    assert torch.allclose(c_final.cpu(), cT_ref, atol=2e-5)
+
+
+@pytest.mark.parametrize("H", [64, 256, 512])
+def test_lstm_fwd_bf16_vs_ref(H):
+    T, B, I = 7, 24, 33
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.bfloat16)
+    h_ref, hT_ref, cT_ref = lstm_seq_ref(*cpu)
+    x, h0, c0, wx, wh, b = gpu
+    gates_x = torch.addmm(b, x.reshape(T * B, I), wx.t()).view(T, B, 4 * H)
+    h_seq, c_final, _, _ = ext().lstm_seq_fwd(gates_x, h0, c0, wh)
+    err = (h_seq.float().cpu() - h_ref).abs().max()
+    assert err < 5e-2, f"bf16 fwd max err {err}"
+
+
+def test_lstm_fwd_ragged_batch():
+    # B not a multiple of 16 exercises the row predicates
+    T, B, I, H = 3, 19, 8, 64
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.float32, seed=3)
+    h_ref, _, cT_ref = lstm_seq_ref(*cpu)
+    x, h0, c0, wx, wh, b = gpu
+    gates_x = torch.addmm(b, x.reshape(T * B, I), wx.t()).view(T, B, 4 * H)
+    h_seq, c_final, _, _ = ext().lstm_seq_fwd(gates_x, h0, c0, wh)
+    assert torch.allclose(h_seq.cpu(), h_ref, atol=2e-5)
+    assert torch.allclose(c_final.cpu(), cT_ref, atol=2e-5)
+
+
+# ------------------------------------------------------------------ K2 bwd
+@pytest.mark.parametrize("H", [64, 256, 512])
+def test_lstm_autograd_f32_vs_ref(H):
+    T, B, I = 5, 17, 12
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.float32, seed=1)
+    cpu = tuple(t.clone().requires_grad_(True) for t in cpu)
+    gpu = tuple(t.clone().requires_grad_(True) for t in gpu)
+
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq
+    h_ref, hT_ref, cT_ref = lstm_seq_ref(*cpu)
+    loss_ref = (h_ref * h_ref).mean() + cT_ref.abs().mean()
+    loss_ref.backward()
+    h_gpu, hT_gpu, cT_gpu = lstm_seq(*gpu)
+    loss_gpu = (h_gpu * h_gpu).mean() + cT_gpu.abs().mean()
+    loss_gpu.backward()
+
+    for name, rc, gc in zip(("x", "h0", "c0", "wx", "wh", "b"), cpu, gpu):
+        err = (rc.grad - gc.grad.cpu()).abs().max()
+        assert err < 1e-4, f"d{name} max err {err}"
+
+
+def test_lstm_autograd_bf16_close():
+    T, B, I, H = 5, 16, 12, 128
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.bfloat16, seed=2)
+    cpu = tuple(t.clone().requires_grad_(True) for t in cpu)
+    gpu = tuple(t.clone().requires_grad_(True) for t in gpu)
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq
+    h_ref, _, cT_ref = lstm_seq_ref(*cpu)
+    (h_ref.square().mean() + cT_ref.abs().mean()).backward()
+    h_gpu, _, cT_gpu = lstm_seq(*gpu)
+    (h_gpu.square().mean() + cT_gpu.abs().mean()).backward()
+    for name, rc, gc in zip(("x", "h0", "c0", "wx", "wh", "b"), cpu, gpu):
+        denom = rc.grad.abs().max().clamp_min(1e-3)
+        err = (rc.grad - gc.grad.float().cpu()).abs().max() / denom
+        assert err < 0.15, f"d{name} rel err {err}"
+
+
+# ---------------------------------------------------------------------- K3
+def test_td_targets_vs_ref():
+    g = torch.Generator().manual_seed(5)
+    B, T, n, gamma, eta = 33, 40, 5, 0.997, 0.9
+    t_ext = T + n
+    r = torch.randn(B, t_ext, generator=g)
+    d = (torch.rand(B, t_ext, generator=g) < 0.05).float()
+    qb = torch.randn(B, T, generator=g)
+    qo = torch.randn(B, T, generator=g)
+    folded, mask = fold_nstep_ref(r, d, gamma, n, T)
+    y_ref = folded + gamma ** n * mask * qb
+    td_ref = (y_ref - qo).abs()
+    p_ref = sequence_priority_ref(td_ref, eta)
+    y, td, p = ext().td_targets(r.cuda(), d.cuda(), qb.cuda(), qo.cuda(), gamma, n, eta)
+    assert torch.allclose(y.cpu(), y_ref, atol=1e-5)
+    assert torch.allclose(td.cpu(), td_ref, atol=1e-5)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+
+
+def test_td_targets_rescale():
+    from pytorch_r2d2_dpg_amd.ops.reference import value_rescale, value_rescale_inv
+    g = torch.Generator().manual_seed(6)
+    B, T, n, gamma, eta = 8, 10, 3, 0.99, 0.9
+    r = torch.randn(B, T + n, generator=g)
+    d = torch.zeros(B, T + n)
+    qb = torch.randn(B, T, generator=g) * 2
+    qo = torch.randn(B, T, generator=g)
+    folded, mask = fold_nstep_ref(r, d, gamma, n, T)
+    y_ref = value_rescale(folded + gamma ** n * mask * value_rescale_inv(qb))
+    y, td, p = ext().td_targets(r.cuda(), d.cuda(), qb.cuda(), qo.cuda(), gamma, n, eta,
+                                rescale=True)
+    assert torch.allclose(y.cpu(), y_ref, atol=1e-4), (y.cpu() - y_ref).abs().max()
+
+
+# ---------------------------------------------------------------------- K4
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_polyak_multi(dtype):
+    from pytorch_r2d2_dpg_amd.ops.polyak import PolyakUpdater
+    g = torch.Generator().manual_seed(7)
+    shapes = [(64, 3), (17,), (128, 128), (5, 7, 3)]
+    tgt = [torch.randn(*s, generator=g).to(dtype).cuda() for s in shapes]
+    onl = [torch.randn(*s, generator=g).to(dtype).cuda() for s in shapes]
+    want = [(1 - 0.01) * t.float() + 0.01 * o.float() for t, o in zip(tgt, onl)]
+    upd = PolyakUpdater(tgt, onl)
+    upd(0.01)
+    tol = 1e-6 if dtype == torch.float32 else 1e-2
+    for t, w in zip(tgt, want):
+        assert torch.allclose(t.float(), w, atol=tol)
+
+
+# ------------------------------------------------------------------- K5/K6
+def test_sumtree_update_matches_cpu_oracle():
+    cap = 1024
+    rng = np.random.default_rng(8)
+    cpu = SumTree(cap)
+    tree = torch.zeros(2 * cpu.capacity, dtype=torch.float32, device="cuda")
+    for _ in range(5):
+        m = 200
+        idx = rng.integers(0, cap, m)
+        # dedupe (duplicate slots in one batch are order-dependent)
+        idx = np.unique(idx)
+        pri = rng.random(idx.size).astype(np.float32) + 0.01
+        cpu.update(idx, pri)
+        ext().sumtree_update(tree, cpu.capacity,
+                             torch.from_numpy(idx.astype(np.int64)).cuda(),
+                             torch.from_numpy(pri).cuda())
+    got = tree.cpu().numpy()
+    assert np.allclose(got[1], cpu.tree[1], rtol=1e-5)
+    assert np.allclose(got[cpu.capacity:], cpu.tree[cpu.capacity:], rtol=1e-5)
+    assert np.allclose(got[1:], cpu.tree[1:], rtol=1e-5)
+
+
+def test_sumtree_sample_distribution_and_determinism():
+    cap = 256
+    rng = np.random.default_rng(9)
+    pri = (rng.random(cap) + 0.05).astype(np.float32)
+    tree = torch.zeros(2 * cap, dtype=torch.float32, device="cuda")
+    ext().sumtree_update(tree, cap, torch.arange(cap, dtype=torch.int64).cuda(),
+                         torch.from_numpy(pri).cuda())
+    counts = np.zeros(cap)
+    B = 512
+    for step in range(200):
+        idx, prob = ext().sumtree_sample(tree, cap, B, 42, step)
+        idx_h = idx.cpu().numpy()
+        counts[idx_h] += 1
+        assert np.allclose(prob.cpu().numpy(), pri[idx_h] / pri.sum(), rtol=1e-4)
+    freq = counts / counts.sum()
+    expect = pri / pri.sum()
+    assert np.abs(freq - expect).max() < 0.003
+    # determinism: same (seed, step) -> same indices
+    i1, _ = ext().sumtree_sample(tree, cap, B, 123, 7)
+    i2, _ = ext().sumtree_sample(tree, cap, B, 123, 7)
+    assert torch.equal(i1, i2)
+    i3, _ = ext().sumtree_sample(tree, cap, B, 123, 8)
+    assert not torch.equal(i1, i3)
+
+
+# ---------------------------------------------------------------------- K7
+def test_fp8_roundtrip_error_bound():
+    g = torch.Generator().manual_seed(10)
+    x = (torch.randn(4096, generator=g) * 3).cuda()
+    enc = torch.empty(4096, dtype=torch.uint8, device="cuda")
+    ext().fp8_encode(x, enc, 1.0)
+    store = enc.view(16, 256)
+    idx = torch.arange(16, dtype=torch.int64).cuda()
+    dec = ext().fp8_gather_decode(store, idx, torch.float32, 1.0).view(-1)
+    # e4m3: 3 mantissa bits -> relative error <= 2^-4 + eps
+    rel = ((dec - x).abs() / x.abs().clamp_min(1e-3)).max()
+    assert rel < 0.07, f"fp8 rel err {rel}"
+
+
+def test_fp8_gather_scrambled_indices():
+    g = torch.Generator().manual_seed(11)
+    x = torch.randn(8, 64, generator=g).cuda()
+    enc = torch.empty_like(x, dtype=torch.uint8)
+    ext().fp8_encode(x.reshape(-1).contiguous(), enc.view(-1), 1.0)
+    idx = torch.tensor([3, 0, 7, 3], dtype=torch.int64).cuda()
+    out = ext().fp8_gather_decode(enc, idx, torch.float32, 1.0)
+    for b, s in enumerate([3, 0, 7, 3]):
+        assert torch.allclose(out[b], x[s], atol=0.25, rtol=0.07)
+
+
+# ------------------------------------------------------- module integration
+def test_gpu_replay_end_to_end():
+    from pytorch_r2d2_dpg_amd.replay.gpu_buffer import GpuSequenceReplay
+    rep = GpuSequenceReplay(capacity=128, stored_len=10, obs_shape=(17,), action_dim=6,
+                            hidden_size=64, device="cuda", seed=0)
+    g = torch.Generator().manual_seed(12)
+    rep.add_batch(torch.randn(32, 10, 17, generator=g), torch.randn(32, 10, 6, generator=g),
+                  torch.rand(32, 10, generator=g), torch.zeros(32, 10),
+                  torch.randn(32, 4, 64, generator=g), torch.rand(32, generator=g) + 0.5)
+    b = rep.sample(16)
+    assert b["obs"].is_cuda and b["obs"].shape == (16, 10, 17)
+    rep.update_priorities(b["idx"], torch.rand(16).cuda() + 0.1)
+    # concentrate priority on one slot -> dominated sampling
+    rep.update_priorities(torch.arange(32).cuda(),
+                          torch.cat([torch.full((1,), 1000.0), torch.full((31,), 1e-6)]).cuda())
+    b2 = rep.sample(64)
+    assert (b2["idx"] == 0).float().mean() > 0.9
+
+
+def test_learner_step_gpu_bf16():
+    from pytorch_r2d2_dpg_amd.config import Config
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+    cfg = Config()
+    cfg.env.name = "humanoid-run"
+    cfg.model.hidden_size = 128
+    cfg.model.trunk_sizes = (128, 128)
+    cfg.model.dtype = "bf16"
+    cfg.replay.seq_len = 16
+    cfg.replay.burn_in = 8
+    cfg.learner.batch_size = 24
+    spec = make_env(cfg.env.name).spec
+    lrn = Learner(cfg, spec, device="cuda")
+    L = cfg.replay.seq_len + cfg.replay.n_step
+    g = torch.Generator().manual_seed(13)
+    batch = {
+        "obs": torch.randn(24, L, 67, generator=g).cuda(),
+        "action": torch.randn(24, L, 21, generator=g).clamp_(-1, 1).cuda(),
+        "reward": torch.rand(24, L, generator=g).cuda(),
+        "done": torch.zeros(24, L).cuda(),
+        "state": torch.zeros(24, 4, 128).cuda(),
+        "weight": torch.ones(24).cuda(),
+        "idx": None,
+    }
+    p0 = [p.detach().clone() for p in lrn.actor.parameters()]
+    stats = lrn.train_step(batch)
+    assert np.isfinite(stats.critic_loss) and np.isfinite(stats.actor_loss)
+    assert any(not torch.equal(a, b) for a, b in zip(p0, lrn.actor.parameters()))

@@ -116,10 +116,21 @@ def main() -> int:
     lrn = Learner(cfg, spec, device=device, dist=dist, replay=replay)
 
     B, T = cfg.learner.batch_size, cfg.train_len
+    lrn.collect_stats = False  # no per-step host syncs in the timed loop
 
     def one_step():
         batch = replay.sample(B)
         lrn.train_step(batch)
+
+    # ---- hipGraph capture of the learner update (launch-bound workload)
+    graphed = False
+    if args.graph and device.startswith("cuda"):
+        try:
+            lrn.capture_graph(replay.sample(B))
+            graphed = True
+        except Exception as e:  # noqa: BLE001
+            print(f"[bench] graph capture failed ({e!r}); running eager",
+                  file=sys.stderr, flush=True)
 
     # ---- warmup
     for _ in range(args.warmup):
@@ -166,6 +177,7 @@ def main() -> int:
                 "n_step": cfg.replay.n_step,
                 "parallelism": f"dp{n_gpus}",
                 "replay": "GPU shard, prioritized (K5/K6), per-rank",
+                "hipgraph": graphed,
             },
         }
         print(json.dumps(out), flush=True)

@@ -10,6 +10,13 @@ Data-parallel: each rank runs this identically on its own shard; gradients
 are all-reduced as one fused bucket over RCCL/xGMI after each backward
 (parallel/dist.py) before the Adam step — DP ranks therefore hold identical
 weights at all times; rank 0 alone checkpoints/logs.
+
+hipGraph capture: the whole update (everything after the replay sample,
+including both backwards, both Adam steps, Polyak and the K6 priority
+write-back) is launch-bound at these model sizes, so ``capture_graph()``
+records it once into a hipGraph and ``train_step`` then just copies the
+sampled batch into the static input buffers and replays — collapsing
+hundreds of launches into one.
 """
 
 from __future__ import annotations
@@ -41,7 +48,7 @@ class StepStats:
 
 class Learner:
     def __init__(self, cfg: Config, spec, device: str = "cpu",
-                 dist: Optional[DistContext] = None, replay: Optional[SequenceReplay] = None):
+                 dist: Optional[DistContext] = None, replay=None):
         self.cfg = cfg
         self.spec = spec
         self.device = torch.device(device)
@@ -69,6 +76,7 @@ class Learner:
 
         self.replay = replay
         self.step_count = 0
+        self.collect_stats = True
 
         R = cfg.replay
         self.burn_in = R.burn_in
@@ -77,19 +85,20 @@ class Learner:
         self.gamma = R.gamma
         self.eta = R.priority_eta
 
-    # ------------------------------------------------------------ train step
-    def train_step(self, batch: Dict[str, torch.Tensor]) -> StepStats:
-        """One learner update on a sampled batch (tensors already on device).
+        self._graph = None
+        self._static: Optional[Dict[str, torch.Tensor]] = None
+        self._graph_stats: Optional[Dict[str, torch.Tensor]] = None
 
-        Batch layout: obs [B, L, ...], action [B, L, A], reward [B, L],
-        done [B, L], state [B, 4, H], weight [B]; L = burn_in + T + n_step.
-        """
+    # ---------------------------------------------------------- core update
+    def _step_core(self, batch: Dict[str, torch.Tensor],
+                   update_priorities: bool = True,
+                   zero_to_none: bool = True) -> Dict[str, torch.Tensor]:
+        """The full device-side update. No host syncs, no Python floats."""
         cfg = self.cfg
-        B = batch["obs"].shape[0]
         Tn = self.train_len + self.n_step
         bi = self.burn_in
+        T = self.train_len
 
-        # time-major views
         obs = batch["obs"].transpose(0, 1).to(self.dtype)          # [L, B, ...]
         action = batch["action"].transpose(0, 1).to(self.dtype)    # [L, B, A]
         reward = batch["reward"]                                   # [B, L] fp32
@@ -111,13 +120,9 @@ class Learner:
         with torch.no_grad():
             mu_t, _ = self.target_actor(obs[bi:bi + Tn], (h_a2, c_a2))
             qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
-            # bootstrap value at t+n for each train-window t: [B, T]
-            q_boot = qt[self.n_step:self.n_step + self.train_len].transpose(0, 1).float()
-            if cfg.learner.value_rescale:
-                q_boot = value_rescale_inv(q_boot)
+            q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
 
         # ---- critic update on n-step targets
-        T = self.train_len
         q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
                                   (h_c2.detach(), c_c2.detach()))
         q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
@@ -125,15 +130,12 @@ class Learner:
             y, td_abs, prio = td_targets_and_priorities(
                 reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),
                 self.gamma, self.n_step, self.eta,
+                rescale=cfg.learner.value_rescale,
             )
-            if cfg.learner.value_rescale:
-                y = value_rescale(y)
-                td_abs = (y - q_online_bt.detach()).abs()
-                prio = self.eta * td_abs.max(1).values + (1 - self.eta) * td_abs.mean(1)
         critic_loss = (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
-        self.opt_critic.zero_grad(set_to_none=True)
+        self.opt_critic.zero_grad(set_to_none=zero_to_none)
         critic_loss.backward()
-        self.dist.allreduce_grads(self.critic.parameters())
+        self.dist.allreduce_grads(self.critic.parameters(), key="critic")
         if cfg.learner.max_grad_norm > 0:
             torch.nn.utils.clip_grad_norm_(self.critic.parameters(), cfg.learner.max_grad_norm)
         self.opt_critic.step()
@@ -142,32 +144,80 @@ class Learner:
         mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
         q_pi, _ = self.critic(obs[bi:bi + T], mu, (h_c2.detach(), c_c2.detach()))
         actor_loss = -q_pi.float().mean()
-        self.opt_actor.zero_grad(set_to_none=True)
-        self.opt_critic.zero_grad(set_to_none=True)  # discard critic grads from this path
+        self.opt_actor.zero_grad(set_to_none=zero_to_none)
+        self.opt_critic.zero_grad(set_to_none=zero_to_none)  # discard this path's grads
         actor_loss.backward()
-        self.dist.allreduce_grads(self.actor.parameters())
+        self.dist.allreduce_grads(self.actor.parameters(), key="actor")
         if cfg.learner.max_grad_norm > 0:
             torch.nn.utils.clip_grad_norm_(self.actor.parameters(), cfg.learner.max_grad_norm)
         self.opt_actor.step()
-        self.opt_critic.zero_grad(set_to_none=True)
+        self.opt_critic.zero_grad(set_to_none=zero_to_none)
 
         # ---- Polyak soft target update (K4: one fused launch per net on GPU)
+        # (branch is Python-side: under graph capture target_update_every == 1)
         if self.step_count % cfg.learner.target_update_every == 0:
             self.polyak_actor(cfg.learner.tau)
             self.polyak_critic(cfg.learner.tau)
 
         # ---- priority write-back (stays on-GPU with the K6 tree kernel)
-        if self.replay is not None and batch.get("idx") is not None:
+        if update_priorities and self.replay is not None and batch.get("idx") is not None:
             self.replay.update_priorities(batch["idx"], prio)
 
+        return {
+            "critic_loss": critic_loss.detach(),
+            "actor_loss": actor_loss.detach(),
+            "mean_q": q_online_bt.detach().mean(),
+            "mean_priority": prio.mean(),
+        }
+
+    # ------------------------------------------------------------ graph cap
+    def capture_graph(self, example_batch: Dict[str, torch.Tensor]) -> None:
+        """Record the whole update into a hipGraph with static input buffers.
+
+        Requires: GPU device, fixed batch shape, target_update_every == 1
+        (the Polyak branch is baked into the recording).
+        """
+        assert self.device.type == "cuda", "graph capture needs a GPU"
+        assert self.cfg.learner.target_update_every == 1
+        keys = ("obs", "action", "reward", "done", "state", "weight", "idx")
+        self._static = {k: example_batch[k].clone() for k in keys}
+        # capturable Adam: state must live on device, step as device tensor
+        for opt in (self.opt_actor, self.opt_critic):
+            for gparam in opt.param_groups:
+                gparam["capturable"] = True
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):  # warmup allocations + optimizer state
+                self._step_core(self._static, zero_to_none=False)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._graph_stats = self._step_core(self._static, zero_to_none=False)
+
+    # ------------------------------------------------------------ train step
+    def train_step(self, batch: Dict[str, torch.Tensor]) -> StepStats:
+        if self._graph is not None:
+            for k, dst in self._static.items():
+                src = batch[k]
+                if src is not None:
+                    dst.copy_(src, non_blocking=True)
+            self._graph.replay()
+            stats = self._graph_stats
+        else:
+            stats = self._step_core(batch)
         self.step_count += 1
-        return StepStats(
-            critic_loss=float(critic_loss.detach()),
-            actor_loss=float(actor_loss.detach()),
-            mean_q=float(q_online_bt.detach().mean()),
-            mean_priority=float(prio.mean()),
-            transitions=B * T,
-        )
+        B = batch["obs"].shape[0]
+        if self.collect_stats:
+            return StepStats(
+                critic_loss=float(stats["critic_loss"]),
+                actor_loss=float(stats["actor_loss"]),
+                mean_q=float(stats["mean_q"]),
+                mean_priority=float(stats["mean_priority"]),
+                transitions=B * self.train_len,
+            )
+        return StepStats(0.0, 0.0, 0.0, 0.0, B * self.train_len)
 
     # -------------------------------------------------------------- weights
     def actor_state_for_sync(self) -> Dict[str, torch.Tensor]:

@@ -20,7 +20,8 @@ from typing import Tuple
 import torch
 
 from . import hip_ext
-from .reference import fold_nstep_ref, sequence_priority_ref
+from .reference import (fold_nstep_ref, sequence_priority_ref, value_rescale,
+                        value_rescale_inv)
 
 
 def td_targets_and_priorities(
@@ -31,6 +32,8 @@ def td_targets_and_priorities(
     gamma: float,
     n_step: int,
     eta: float,
+    rescale: bool = False,        # R2D2 value rescaling h(.) (flag, default off)
+    rescale_eps: float = 1e-3,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Returns (targets y [B,T], |delta| [B,T], priorities [B])."""
     T = q_online.shape[1]
@@ -38,9 +41,13 @@ def td_targets_and_priorities(
         return hip_ext().td_targets(
             rewards.contiguous(), dones.contiguous(), bootstrap_q.contiguous(),
             q_online.contiguous(), float(gamma), int(n_step), float(eta),
+            rescale, rescale_eps,
         )
+    qb = value_rescale_inv(bootstrap_q, rescale_eps) if rescale else bootstrap_q
     folded, mask = fold_nstep_ref(rewards, dones, gamma, n_step, T)
-    y = folded + (gamma ** n_step) * mask * bootstrap_q
+    y = folded + (gamma ** n_step) * mask * qb
+    if rescale:
+        y = value_rescale(y, rescale_eps)
     td_abs = (y - q_online).abs()
     prio = sequence_priority_ref(td_abs, eta)
     return y, td_abs, prio

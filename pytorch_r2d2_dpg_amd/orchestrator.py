@@ -1,0 +1,113 @@
+"""Node orchestrator.
+
+Wires the full distributed run on one node: k learner ranks (one per GPU,
+launched via torchrun / torch.distributed.run), each owning a GPU replay
+shard and a slice of the CPU actor pool. Ingest is drained between learner
+steps (single-writer contract); weights are published to actors through the
+shared-memory buffer every sync_every steps; rank 0 checkpoints and logs.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Optional
+
+import numpy as np
+import torch
+
+from .checkpoint import load_checkpoint, save_checkpoint
+from .config import Config
+from .envs import make_env
+from .learner import Learner
+from .parallel.actor_pool import ActorPool
+from .parallel.dist import DistContext
+from .replay.gpu_buffer import GpuSequenceReplay
+from .utils.metrics import MetricsLogger
+
+
+def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str] = None,
+                 log_path: Optional[str] = None, max_steps: Optional[int] = None):
+    dist = DistContext.from_env()
+    rank, world = dist.rank, dist.world_size
+    if device is None:
+        if torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", rank))
+            device = f"cuda:{local}"
+            torch.cuda.set_device(device)
+        else:
+            device = "cpu"
+
+    spec = make_env(cfg.env.name).spec
+    R = cfg.replay
+    replay = GpuSequenceReplay(
+        capacity=R.capacity, stored_len=R.seq_len + R.n_step,
+        obs_shape=spec.obs_shape, action_dim=spec.action_dim,
+        hidden_size=cfg.model.hidden_size, per_alpha=R.per_alpha, per_beta=R.per_beta,
+        device=device, fp8_obs=R.fp8_obs, seed=cfg.learner.seed + 1000 * rank,
+        obs_dtype=torch.uint8 if spec.pixel else torch.float32,
+    )
+    learner = Learner(cfg, spec, device=device, dist=dist, replay=replay)
+    if resume:
+        load_checkpoint(learner, resume)
+
+    actors_per_rank = max(1, cfg.actor.num_actors // world)
+    pool = ActorPool(cfg, actors_per_rank, base_actor_id=rank * actors_per_rank)
+    pool.start()
+    logger = MetricsLogger(log_path, stdout_every=1, rank=rank)
+
+    def ingest():
+        seqs = pool.poll()
+        if seqs:
+            obs = torch.from_numpy(np.stack([s.obs for s in seqs]))
+            act = torch.from_numpy(np.stack([s.action for s in seqs]))
+            rew = torch.from_numpy(np.stack([s.reward for s in seqs]))
+            done = torch.from_numpy(np.stack([s.done for s in seqs]))
+            st = torch.from_numpy(np.stack([s.state for s in seqs]))
+            prio = torch.tensor([s.priority for s in seqs], dtype=torch.float32)
+            replay.add_batch(obs, act, rew, done, st, prio)
+        return len(seqs)
+
+    steps = max_steps if max_steps is not None else cfg.learner.steps
+    try:
+        # ---- warm the replay
+        while len(replay) < max(R.min_fill, cfg.learner.batch_size):
+            ingest()
+            pool.watchdog()
+            time.sleep(0.05)
+
+        t_last = time.time()
+        trans_acc = 0
+        env_steps_last = pool.aggregate_env_steps()
+        for step in range(learner.step_count, steps):
+            ingest()
+            batch = replay.sample(cfg.learner.batch_size)
+            stats = learner.train_step(batch)
+            trans_acc += stats.transitions
+            if step % cfg.learner.sync_every == 0:
+                sd = learner.actor_state_for_sync()
+                pool.publish_weights(sd["actor"], sd["critic"])
+                pool.watchdog()
+            if rank == 0 and cfg.learner.checkpoint_every and \
+                    (step + 1) % cfg.learner.checkpoint_every == 0:
+                save_checkpoint(learner, cfg.learner.checkpoint_dir)
+            if step % max(cfg.learner.log_every, 1) == 0:
+                now = time.time()
+                dt = max(now - t_last, 1e-9)
+                env_steps = pool.aggregate_env_steps()
+                recent = [r for _, r in pool.episode_returns[-20:]]
+                logger.log(
+                    step,
+                    critic_loss=stats.critic_loss, actor_loss=stats.actor_loss,
+                    mean_q=stats.mean_q,
+                    learner_transitions_per_s=trans_acc / dt,
+                    actor_fps=(env_steps - env_steps_last) / dt,
+                    replay_size=len(replay), replay_added=replay.total_added,
+                    episode_return=float(np.mean(recent)) if recent else 0.0,
+                    actor_respawns=pool.respawn_count,
+                )
+                t_last, trans_acc, env_steps_last = now, 0, env_steps
+    finally:
+        pool.stop()
+        logger.close()
+    return learner

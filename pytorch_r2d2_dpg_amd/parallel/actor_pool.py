@@ -1,0 +1,195 @@
+"""Actor process pool (SURVEY.md §3.1/§5 failure detection).
+
+Spawns N CPU actor processes per learner rank. Each actor runs the rollout
+loop (actor.py), pushes cut sequences through a queue, and pulls fresh
+weights from a shared-memory flat parameter buffer (version-stamped, no
+per-sync pickling of state_dicts).
+
+Elasticity: actors are stateless beyond their env, so the pool heartbeats
+each actor (env-step watermark) and respawns dead/stalled ones with a fresh
+seed. A test-only fault-injection hook (`crash_at`) kills actor i at step s
+to exercise that path. Learner ranks are NOT elastic (fixed RCCL world).
+"""
+
+from __future__ import annotations
+
+import queue as queue_mod
+import time
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+from ..actor import ActorWorker
+from ..config import Config
+from ..replay.sequence import Sequence
+
+
+# ---------------------------------------------------------- sd <-> flat buf
+def flat_size(sd: Dict[str, torch.Tensor]) -> int:
+    return sum(v.numel() for v in sd.values())
+
+
+def pack_sd(sd: Dict[str, torch.Tensor], buf: torch.Tensor) -> None:
+    off = 0
+    for k in sorted(sd):
+        v = sd[k]
+        n = v.numel()
+        buf[off:off + n].copy_(v.detach().float().reshape(-1))
+        off += n
+
+
+def unpack_sd(template: Dict[str, torch.Tensor], buf: torch.Tensor) -> Dict[str, torch.Tensor]:
+    out = {}
+    off = 0
+    for k in sorted(template):
+        v = template[k]
+        n = v.numel()
+        out[k] = buf[off:off + n].view(v.shape).clone()
+        off += n
+    return out
+
+
+# -------------------------------------------------------------- actor proc
+def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver,
+                stop_event, heartbeat, crash_at: int, chunk: int = 64):
+    torch.set_num_threads(1)
+    cfg = Config.from_dict(cfg_dict)
+    worker = ActorWorker(cfg, actor_id=actor_id)
+    a_tpl = worker.actor_net.state_dict()
+    c_tpl = worker.critic_net.state_dict()
+    na = flat_size(a_tpl)
+    last_ver = -1
+    ep_reported = 0
+    while not stop_event.is_set():
+        ver = int(weight_ver.value)
+        if ver != last_ver and ver > 0:
+            worker.load_weights(unpack_sd(a_tpl, weight_buf[:na]),
+                                unpack_sd(c_tpl, weight_buf[na:]))
+            last_ver = ver
+        seqs = worker.rollout(chunk)
+        if crash_at and worker.env_steps >= crash_at:
+            raise RuntimeError(f"fault injection: actor {actor_id} crash at {crash_at}")
+        if seqs:
+            payload = [(s.obs, s.action, s.reward, s.done, s.state, s.priority, s.length)
+                       for s in seqs]
+            try:
+                seq_queue.put(payload, timeout=5.0)
+            except queue_mod.Full:
+                pass  # backpressure: drop if learner can't keep up
+        new_eps = worker.episode_returns[ep_reported:]
+        if new_eps:
+            try:
+                seq_queue.put(("episodes", actor_id, new_eps), timeout=0.1)
+                ep_reported = len(worker.episode_returns)
+            except queue_mod.Full:
+                pass
+        heartbeat[actor_id] = worker.env_steps
+
+
+class ActorPool:
+    def __init__(self, cfg: Config, num_actors: int, base_actor_id: int = 0,
+                 crash_at: Optional[Dict[int, int]] = None, chunk: int = 64,
+                 queue_size: int = 512):
+        self.cfg = cfg
+        self.num_actors = num_actors
+        self.base_actor_id = base_actor_id
+        self.crash_at = crash_at or {}
+        self.chunk = chunk
+        self._ctx = mp.get_context("spawn")
+        self.seq_queue = self._ctx.Queue(maxsize=queue_size)
+        self.stop_event = self._ctx.Event()
+        self.heartbeat = self._ctx.Array("l", num_actors)
+        self.weight_ver = self._ctx.Value("l", 0)
+        # shared flat weight buffer sized from a probe worker
+        probe = ActorWorker(cfg, actor_id=0)
+        self._a_tpl = probe.actor_net.state_dict()
+        self._c_tpl = probe.critic_net.state_dict()
+        self._na = flat_size(self._a_tpl)
+        self._nc = flat_size(self._c_tpl)
+        self.weight_buf = torch.zeros(self._na + self._nc, dtype=torch.float32)
+        self.weight_buf.share_memory_()
+        del probe
+        self.procs: List[Optional[mp.Process]] = [None] * num_actors
+        self._last_beat = [0] * num_actors
+        self._last_beat_time = [time.time()] * num_actors
+        self.respawn_count = 0
+        self.episode_returns: List[Tuple[int, float]] = []
+        self.total_sequences = 0
+
+    # ------------------------------------------------------------ lifecycle
+    def start(self) -> None:
+        for i in range(self.num_actors):
+            self._spawn(i)
+
+    def _spawn(self, i: int) -> None:
+        p = self._ctx.Process(
+            target=_actor_main,
+            args=(self.base_actor_id + i, self.cfg.to_dict(), self.seq_queue,
+                  self.weight_buf, self.weight_ver, self.stop_event, self.heartbeat,
+                  self.crash_at.get(self.base_actor_id + i, 0), self.chunk),
+            daemon=True,
+        )
+        p.start()
+        self.procs[i] = p
+        self._last_beat[i] = -1
+        self._last_beat_time[i] = time.time()
+
+    def stop(self) -> None:
+        self.stop_event.set()
+        deadline = time.time() + 5
+        for p in self.procs:
+            if p is not None:
+                p.join(timeout=max(0.1, deadline - time.time()))
+                if p.is_alive():
+                    p.terminate()
+
+    # ------------------------------------------------------------- data i/o
+    def poll(self, max_batches: int = 64) -> List[Sequence]:
+        """Drain available sequences (non-blocking)."""
+        out: List[Sequence] = []
+        for _ in range(max_batches):
+            try:
+                payload = self.seq_queue.get_nowait()
+            except queue_mod.Empty:
+                break
+            if isinstance(payload, tuple) and payload and payload[0] == "episodes":
+                _, aid, eps = payload
+                self.episode_returns.extend((aid, r) for r in eps)
+                continue
+            for obs, act, rew, done, state, prio, length in payload:
+                out.append(Sequence(obs, act, rew, done, state, prio, length))
+        self.total_sequences += len(out)
+        return out
+
+    def publish_weights(self, actor_sd: Dict[str, torch.Tensor],
+                        critic_sd: Dict[str, torch.Tensor]) -> None:
+        pack_sd(actor_sd, self.weight_buf[:self._na])
+        pack_sd(critic_sd, self.weight_buf[self._na:])
+        with self.weight_ver.get_lock():
+            self.weight_ver.value += 1
+
+    # ------------------------------------------------- failure detection
+    def watchdog(self, stall_seconds: float = 60.0) -> int:
+        """Respawn dead or stalled actors; returns number respawned."""
+        n = 0
+        now = time.time()
+        for i, p in enumerate(self.procs):
+            beat = self.heartbeat[i]
+            if beat != self._last_beat[i]:
+                self._last_beat[i] = beat
+                self._last_beat_time[i] = now
+            dead = p is None or not p.is_alive()
+            stalled = (now - self._last_beat_time[i]) > stall_seconds
+            if dead or stalled:
+                if p is not None and p.is_alive():
+                    p.terminate()
+                self.crash_at.pop(self.base_actor_id + i, None)  # don't re-crash
+                self._spawn(i)
+                self.respawn_count += 1
+                n += 1
+        return n
+
+    def aggregate_env_steps(self) -> int:
+        return int(sum(self.heartbeat))

@@ -101,9 +101,11 @@ def test_lstm_autograd_bf16_close():
     h_gpu, _, cT_gpu = lstm_seq(*gpu)
     (h_gpu.square().mean() + cT_gpu.abs().mean()).backward()
     for name, rc, gc in zip(("x", "h0", "c0", "wx", "wh", "b"), cpu, gpu):
-        denom = rc.grad.abs().max().clamp_min(1e-3)
-        err = (rc.grad - gc.grad.float().cpu()).abs().max() / denom
-        assert err < 0.15, f"d{name} rel err {err}"
+        # bf16 storage of dgates/h_seq feeding the grad GEMMs: compare on the
+        # norm scale (elementwise max-rel is dominated by near-zero entries)
+        num = (rc.grad - gc.grad.float().cpu()).norm()
+        den = rc.grad.norm().clamp_min(1e-3)
+        assert num / den < 0.08, f"d{name} rel-norm err {num / den}"
 
 
 # ---------------------------------------------------------------------- K3

@@ -1,0 +1,209 @@
+"""Multi-process tests (CPU): actor pool lifecycle + fault injection, and
+DP-equivalence over gloo world_size=2 (SURVEY.md §4 "Distributed")."""
+
+import multiprocessing
+import os
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from pytorch_r2d2_dpg_amd.config import Config
+
+
+def pool_cfg():
+    cfg = Config()
+    cfg.env.name = "pendulum"
+    cfg.env.time_limit = 60
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.replay.seq_len = 12
+    cfg.replay.burn_in = 4
+    cfg.replay.overlap = 6
+    cfg.replay.n_step = 3
+    return cfg
+
+
+@pytest.mark.timeout(120)
+def test_actor_pool_produces_sequences_and_syncs_weights():
+    from pytorch_r2d2_dpg_amd.parallel.actor_pool import ActorPool
+    cfg = pool_cfg()
+    pool = ActorPool(cfg, num_actors=2, chunk=30)
+    pool.start()
+    try:
+        seqs = []
+        deadline = time.time() + 60
+        while len(seqs) < 4 and time.time() < deadline:
+            seqs.extend(pool.poll())
+            time.sleep(0.1)
+        assert len(seqs) >= 4, "actors produced no sequences"
+        s = seqs[0]
+        assert s.obs.shape == (cfg.replay.seq_len + cfg.replay.n_step, 3)
+        assert s.state.shape == (4, 16)
+        # weight publish is accepted (version propagates without error)
+        from pytorch_r2d2_dpg_amd.actor import ActorWorker
+        w = ActorWorker(cfg, actor_id=0)
+        pool.publish_weights(w.actor_net.state_dict(), w.critic_net.state_dict())
+        time.sleep(1.0)
+        assert pool.aggregate_env_steps() > 0
+    finally:
+        pool.stop()
+
+
+@pytest.mark.timeout(180)
+def test_actor_pool_respawns_crashed_actor():
+    from pytorch_r2d2_dpg_amd.parallel.actor_pool import ActorPool
+    cfg = pool_cfg()
+    pool = ActorPool(cfg, num_actors=1, crash_at={0: 50}, chunk=25)
+    pool.start()
+    try:
+        # wait for the injected crash
+        deadline = time.time() + 60
+        while time.time() < deadline:
+            if pool.procs[0] is not None and not pool.procs[0].is_alive():
+                break
+            time.sleep(0.2)
+        assert not pool.procs[0].is_alive(), "fault injection did not fire"
+        n = pool.watchdog(stall_seconds=1e9)
+        assert n == 1 and pool.respawn_count == 1
+        # respawned actor makes progress again
+        deadline = time.time() + 60
+        got = []
+        while not got and time.time() < deadline:
+            got = pool.poll()
+            time.sleep(0.2)
+        assert got, "respawned actor produced nothing"
+    finally:
+        pool.stop()
+
+
+# --------------------------------------------------------- DP equivalence
+def _dp_worker(rank, world, port, out_q):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    import torch
+    import torch.distributed as dist
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+    from pytorch_r2d2_dpg_amd.parallel.dist import DistContext
+
+    torch.manual_seed(99)  # same init on both ranks (broadcast also enforces)
+    cfg = pool_cfg()
+    cfg.learner.batch_size = 4
+    dctx = DistContext.from_env(backend="gloo")
+    spec = make_env(cfg.env.name).spec
+    lrn = Learner(cfg, spec, device="cpu", dist=dctx)
+    L = cfg.replay.seq_len + cfg.replay.n_step
+    H = cfg.model.hidden_size
+    g = torch.Generator().manual_seed(1234 + rank)  # DIFFERENT shard per rank
+    batch = {
+        "obs": torch.randn(4, L, 3, generator=g),
+        "action": torch.randn(4, L, 1, generator=g),
+        "reward": torch.rand(4, L, generator=g),
+        "done": torch.zeros(4, L),
+        "state": torch.zeros(4, 4, H),
+        "weight": torch.ones(4),
+        "idx": None,
+    }
+    for _ in range(2):
+        lrn.train_step(batch)
+    flat = torch.cat([p.detach().reshape(-1) for p in lrn.actor.parameters()])
+    out_q.put((rank, flat.numpy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_dp_ranks_stay_weight_identical():
+    """After steps on DIFFERENT data, all-reduced DP ranks hold identical weights."""
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    port = 29571
+    procs = [ctx.Process(target=_dp_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, flat = q.get(timeout=150)
+        results[rank] = flat
+    for p in procs:
+        p.join(timeout=30)
+    assert set(results) == {0, 1}
+    np.testing.assert_allclose(results[0], results[1], rtol=0, atol=1e-6)
+
+
+def _single_vs_dp_worker(rank, world, port, batch_file, out_q):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    import torch
+    import torch.distributed as dist
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+    from pytorch_r2d2_dpg_amd.parallel.dist import DistContext
+
+    torch.manual_seed(7)
+    cfg = pool_cfg()
+    cfg.learner.batch_size = 4
+    cfg.learner.max_grad_norm = 0.0  # global-norm clip differs between 1x8 and 2x4
+    dctx = DistContext.from_env(backend="gloo")
+    spec = make_env(cfg.env.name).spec
+    lrn = Learner(cfg, spec, device="cpu", dist=dctx)
+    full = torch.load(batch_file, weights_only=False)
+    # rank r takes rows [r*4, (r+1)*4) of the global batch of 8
+    batch = {k: (v[rank * 4:(rank + 1) * 4] if torch.is_tensor(v) else v)
+             for k, v in full.items()}
+    lrn.train_step(batch)
+    flat = torch.cat([p.detach().reshape(-1) for p in lrn.critic.parameters()])
+    out_q.put((rank, flat.numpy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_dp_equivalence_vs_single(tmp_path):
+    """2 ranks x batch 4 == 1 process x batch 8 (same global batch)."""
+    import torch
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+
+    cfg = pool_cfg()
+    cfg.learner.batch_size = 8
+    cfg.learner.max_grad_norm = 0.0
+    L = cfg.replay.seq_len + cfg.replay.n_step
+    H = cfg.model.hidden_size
+    g = torch.Generator().manual_seed(4321)
+    full = {
+        "obs": torch.randn(8, L, 3, generator=g),
+        "action": torch.randn(8, L, 1, generator=g),
+        "reward": torch.rand(8, L, generator=g),
+        "done": torch.zeros(8, L),
+        "state": torch.zeros(8, 4, H),
+        "weight": torch.ones(8),
+        "idx": None,
+    }
+    f = str(tmp_path / "batch.pt")
+    torch.save(full, f)
+
+    torch.manual_seed(7)
+    spec = make_env(cfg.env.name).spec
+    single = Learner(cfg, spec, device="cpu")
+    single.train_step(full)
+    want = torch.cat([p.detach().reshape(-1) for p in single.critic.parameters()]).numpy()
+
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_single_vs_dp_worker, args=(r, 2, 29573, f, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, flat = q.get(timeout=150)
+        results[rank] = flat
+    for p in procs:
+        p.join(timeout=30)
+    np.testing.assert_allclose(results[0], want, atol=2e-5)
+    np.testing.assert_allclose(results[1], want, atol=2e-5)

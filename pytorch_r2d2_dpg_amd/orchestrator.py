@@ -23,6 +23,7 @@ from .learner import Learner
 from .parallel.actor_pool import ActorPool
 from .parallel.dist import DistContext
 from .replay.gpu_buffer import GpuSequenceReplay
+from .replay.staging import StagingRing
 from .utils.metrics import MetricsLogger
 
 
@@ -56,16 +57,12 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
     pool.start()
     logger = MetricsLogger(log_path, stdout_every=1, rank=rank)
 
+    staging = StagingRing(replay)
+
     def ingest():
         seqs = pool.poll()
         if seqs:
-            obs = torch.from_numpy(np.stack([s.obs for s in seqs]))
-            act = torch.from_numpy(np.stack([s.action for s in seqs]))
-            rew = torch.from_numpy(np.stack([s.reward for s in seqs]))
-            done = torch.from_numpy(np.stack([s.done for s in seqs]))
-            st = torch.from_numpy(np.stack([s.state for s in seqs]))
-            prio = torch.tensor([s.priority for s in seqs], dtype=torch.float32)
-            replay.add_batch(obs, act, rew, done, st, prio)
+            staging.ingest(seqs)
         return len(seqs)
 
     steps = max_steps if max_steps is not None else cfg.learner.steps

@@ -1,0 +1,91 @@
+"""Pinned-host staging ring for actor->GPU replay ingest.
+
+Actors produce numpy sequences on CPU; the learner-side ingest packs them
+into a ring of pinned host buffers and issues the H2D copy on a dedicated
+side stream (hipMemcpyAsync under torch), so the bulk bytes overlap the
+learner's compute. The scatter into the shard + sum-tree publish then run on
+the learner's stream after an event wait, preserving the single-writer /
+publish-last ordering (§3.4) without ever blocking on the copy.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .gpu_buffer import GpuSequenceReplay
+from .sequence import Sequence
+
+
+class StagingRing:
+    def __init__(self, replay: GpuSequenceReplay, slots: int = 4, slot_seqs: int = 128):
+        self.replay = replay
+        self.device = replay.device
+        self.is_cuda = replay.is_cuda
+        self.slot_seqs = slot_seqs
+        self.slots = slots
+        L = replay.stored_len
+        obs_shape = tuple(replay.obs.shape[2:])
+        A = replay.action.shape[2]
+        H = replay.state.shape[2]
+        pin = self.is_cuda
+
+        def mk(shape, dtype=torch.float32):
+            return [torch.empty((slot_seqs,) + shape, dtype=dtype, pin_memory=pin)
+                    for _ in range(slots)]
+
+        # obs staged in fp32 (fp8 encode happens on-GPU at ingest, K7)
+        self._obs = mk((L,) + obs_shape,
+                       torch.float32 if replay.fp8_obs else replay.obs.dtype)
+        self._act = mk((L, A))
+        self._rew = mk((L,))
+        self._done = mk((L,))
+        self._state = mk((4, H))
+        self._prio = mk(())
+        self._events = [torch.cuda.Event() if pin else None for _ in range(slots)]
+        self._busy = [False] * slots
+        self._cursor = 0
+        self.stream = torch.cuda.Stream() if pin else None
+
+    def _next_slot(self) -> int:
+        s = self._cursor
+        if self._busy[s]:
+            self._events[s].synchronize()  # ring full: wait for oldest copy
+            self._busy[s] = False
+        self._cursor = (s + 1) % self.slots
+        return s
+
+    @torch.no_grad()
+    def ingest(self, seqs: List[Sequence]) -> int:
+        """Stage and enqueue sequences into the replay shard. Returns count."""
+        total = 0
+        for off in range(0, len(seqs), self.slot_seqs):
+            chunk = seqs[off:off + self.slot_seqs]
+            n = len(chunk)
+            s = self._next_slot()
+            obs_t = self._obs[s][:n]
+            obs_np = np.stack([q.obs for q in chunk])
+            obs_t.copy_(torch.from_numpy(obs_np).to(obs_t.dtype))
+            self._act[s][:n].copy_(torch.from_numpy(np.stack([q.action for q in chunk])))
+            self._rew[s][:n].copy_(torch.from_numpy(np.stack([q.reward for q in chunk])))
+            self._done[s][:n].copy_(torch.from_numpy(np.stack([q.done for q in chunk])))
+            self._state[s][:n].copy_(torch.from_numpy(np.stack([q.state for q in chunk])))
+            self._prio[s][:n].copy_(torch.tensor([q.priority for q in chunk]))
+            if self.is_cuda:
+                # bulk H2D on the side stream, scatter on the learner stream
+                with torch.cuda.stream(self.stream):
+                    dev = [t[:n].to(self.device, non_blocking=True)
+                           for t in (self._obs[s], self._act[s], self._rew[s],
+                                     self._done[s], self._state[s], self._prio[s])]
+                    self._events[s].record(self.stream)
+                self._busy[s] = True
+                torch.cuda.current_stream().wait_event(self._events[s])
+                self.replay.add_batch(*dev)
+            else:
+                self.replay.add_batch(self._obs[s][:n], self._act[s][:n],
+                                      self._rew[s][:n], self._done[s][:n],
+                                      self._state[s][:n], self._prio[s][:n])
+            total += n
+        return total

@@ -214,9 +214,11 @@ def test_fp8_roundtrip_error_bound():
     store = enc.view(16, 256)
     idx = torch.arange(16, dtype=torch.int64).cuda()
     dec = ext().fp8_gather_decode(store, idx, torch.float32, 1.0).view(-1)
-    # e4m3: 3 mantissa bits -> relative error <= 2^-4 + eps
-    rel = ((dec - x).abs() / x.abs().clamp_min(1e-3)).max()
-    assert rel < 0.07, f"fp8 rel err {rel}"
+    # e4m3: 3 mantissa bits -> rel err <= 2^-4 in the normal range, absolute
+    # quantum 2^-9 in the subnormal range near zero
+    bound = x.abs() / 15.0 + 2.0 ** -9
+    assert ((dec - x).abs() <= bound).all(), \
+        f"fp8 err {(dec - x).abs().max()} worst at {x[(dec - x).abs().argmax()]}"
 
 
 def test_fp8_gather_scrambled_indices():

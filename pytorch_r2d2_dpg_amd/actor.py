@@ -81,7 +81,8 @@ class ActorWorker:
             next_obs, r, done, trunc = self.env.step(a)
             self.episode_return += r
             self.env_steps += 1
-            out.extend(self.cutter.push(self._obs, a, r, done, float(q[0]), snapshot))
+            out.extend(self.cutter.push(self._obs, a, r * self.cfg.replay.reward_scale,
+                                        done, float(q[0]), snapshot))
             self._obs = next_obs
             if done or trunc:
                 if trunc and not done:

@@ -52,6 +52,7 @@ class ReplayConfig:
     per_alpha: float = 0.9
     per_beta: float = 0.0           # importance-sampling exponent (R2D2 used 0)
     min_fill: int = 256             # sequences before learning starts
+    reward_scale: float = 1.0       # rewards are scaled at actor-push time
     fp8_obs: bool = False           # store observations as fp8 e4m3 in the GPU shard
     device: str = "auto"            # auto: GPU shard when learner is on GPU
 
@@ -62,6 +63,7 @@ class LearnerConfig:
     lr_actor: float = 1e-4
     lr_critic: float = 1e-4
     adam_eps: float = 1e-8
+    critic_weight_decay: float = 0.0  # DDPG-paper L2 on the critic
     tau: float = 5e-3               # Polyak soft-update rate
     target_update_every: int = 1
     value_rescale: bool = False     # h(x)=sign(x)(sqrt(|x|+1)-1)+eps*x (R2D2-DQN detail, default off)

@@ -68,7 +68,8 @@ class Learner:
 
         lc = cfg.learner
         self.opt_actor = torch.optim.Adam(self.actor.parameters(), lr=lc.lr_actor, eps=lc.adam_eps)
-        self.opt_critic = torch.optim.Adam(self.critic.parameters(), lr=lc.lr_critic, eps=lc.adam_eps)
+        self.opt_critic = torch.optim.Adam(self.critic.parameters(), lr=lc.lr_critic, eps=lc.adam_eps,
+                                           weight_decay=lc.critic_weight_decay)
         self.polyak_actor = PolyakUpdater(
             list(self.target_actor.parameters()), list(self.actor.parameters()))
         self.polyak_critic = PolyakUpdater(
@@ -99,7 +100,9 @@ class Learner:
         bi = self.burn_in
         T = self.train_len
 
-        obs = batch["obs"].transpose(0, 1).to(self.dtype)          # [L, B, ...]
+        obs = batch["obs"].transpose(0, 1)                         # [L, B, ...]
+        if obs.dtype != torch.uint8:  # pixel obs stay uint8 (trunk normalizes)
+            obs = obs.to(self.dtype)
         action = batch["action"].transpose(0, 1).to(self.dtype)    # [L, B, A]
         reward = batch["reward"]                                   # [B, L] fp32
         done = batch["done"]                                       # [B, L] fp32

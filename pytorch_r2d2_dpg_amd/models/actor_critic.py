@@ -45,9 +45,12 @@ class ConvTrunk(nn.Module):
             self.out_dim = self.net(torch.zeros(1, 3, 84, 84)).shape[1]
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        # x: [N, 84, 84, 3] uint8 or float -> [N, out_dim]
+        # x: [N, 84, 84, 3] uint8 (raw) or float (pre-normalized) -> [N, out_dim]
+        wdtype = self.net[0].weight.dtype
         if x.dtype == torch.uint8:
-            x = x.to(torch.float32).div_(255.0)
+            x = x.to(wdtype) / 255.0
+        else:
+            x = x.to(wdtype)
         return self.net(x.permute(0, 3, 1, 2).contiguous())
 
 
@@ -83,6 +86,12 @@ class ActorNet(nn.Module):
         self.trunk = _TrunkEncoder(obs_shape, trunk_sizes, conv_channels, pixel)
         self.lstm = FusedLSTM(self.trunk.out_dim, hidden_size)
         self.head = nn.Linear(hidden_size, action_dim)
+        # DDPG-paper head init (+-3e-3): keeps initial actions near 0 so the
+        # tanh does not saturate before the critic is informative (saturated
+        # tanh kills the actor gradient permanently)
+        with torch.no_grad():
+            self.head.weight.uniform_(-3e-3, 3e-3)
+            self.head.bias.zero_()
 
     def zero_state(self, batch: int, device=None, dtype=None) -> State:
         return self.lstm.zero_state(batch, device, dtype)
@@ -108,6 +117,9 @@ class CriticNet(nn.Module):
         self.trunk = _TrunkEncoder(obs_shape, trunk_sizes, conv_channels, pixel)
         self.lstm = FusedLSTM(self.trunk.out_dim + action_dim, hidden_size)
         self.head = nn.Linear(hidden_size, 1)
+        with torch.no_grad():
+            self.head.weight.uniform_(-3e-3, 3e-3)
+            self.head.bias.zero_()
 
     def zero_state(self, batch: int, device=None, dtype=None) -> State:
         return self.lstm.zero_state(batch, device, dtype)

@@ -31,8 +31,15 @@
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-DEV_INLINE float sigm(float x) { return 1.0f / (1.0f + __expf(-x)); }
-DEV_INLINE float tanh_e(float x) { return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f; }
+// Hardware transcendentals: v_exp_f32 (2^x) + v_rcp_f32. Plain `1.0f/x`
+// emits the IEEE div_scale/div_fmas/div_fixup sequence (~10 ops each, 40
+// per wave-step here — measured as the epilogue's dominant cost).
+DEV_INLINE float sigm(float x) {
+    return __builtin_amdgcn_rcpf(1.0f + __builtin_amdgcn_exp2f(-1.44269504f * x));
+}
+DEV_INLINE float tanh_e(float x) {
+    return 2.0f * __builtin_amdgcn_rcpf(1.0f + __builtin_amdgcn_exp2f(-2.88539008f * x)) - 1.0f;
+}
 
 // XOR swizzle: spread 16 rows of a row-major LDS tile over bank slots so the
 // 16-lane groups of ds_read_b128 hit distinct 16B slots (<=2-way).
@@ -70,7 +77,7 @@ DEV_INLINE void st_lds(char* smem, unsigned byte_off, float v) {
 
 // ---------------------------------------------------------------- forward
 template <typename DT, int HFRAG>
-__global__ void __launch_bounds__(512)
+__global__ void __launch_bounds__(HFRAG <= 2 ? 1024 : 512)
 lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                 const DT* __restrict__ h0,       // [B, H]
                 const DT* __restrict__ c0,       // [B, H]
@@ -81,7 +88,9 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                 float* __restrict__ c_seq,       // [T, B, H]
                 int T, int B, int H) {
     using F = Frag<DT>;
-    extern __shared__ __attribute__((aligned(16))) char smem[];  // h tile [16][H] DT, swizzled
+    // TWO h tiles (double buffer): read h(t-1) from buf[t&1], write h(t) to
+    // buf[(t+1)&1] -> one barrier per timestep instead of two.
+    extern __shared__ __attribute__((aligned(16))) char smem[];
 
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
@@ -97,7 +106,8 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
     const int c_col = lane & 15;              // C/D col within fragment
     const int c_row0 = (lane >> 4) * 4;       // C/D first row (4 regs)
 
-    // ---- load h0 tile into LDS (swizzled), c0 into C-layout registers
+    const unsigned tile_bytes = 16 * rowbytes;
+    // ---- load h0 tile into LDS buf[0] (swizzled), c0 into C-layout registers
     for (int e = tid; e < 16 * H; e += (int)blockDim.x) {
         int r = e / H, col = e - r * H;
         int grow = brow0 + r;
@@ -120,7 +130,7 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
     // chunk's epilogue — at H=512 (HFRAG=8) a monolithic acc[4][8] spills.
     // The new h values are held in registers and written to LDS only after
     // EVERY chunk has finished reading h(t-1).
-    constexpr int CHUNK = (HFRAG > 4 && HFRAG % 4 == 0) ? 4 : HFRAG;
+    constexpr int CHUNK = (HFRAG > 2 && HFRAG % 2 == 0) ? 2 : HFRAG;
     constexpr int NCHUNK = HFRAG / CHUNK;
 
     // ---- time loop
@@ -144,20 +154,29 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                         int j = g * H + col_base + (cc * CHUNK + mc) * 16 + c_col;
                         acc[g][mc][r] = (grow < B) ? ld_f(&gx_t[(long)grow * H4 + j]) : 0.0f;
                     }
-            // K-loop over H: A from LDS (h tile), B from global (wh, L2-resident)
+            // K-loop over H: A from LDS (h tile), B from global (wh,
+            // L2-resident). Two-phase body: issue ALL of this k-step's B
+            // fragment loads back-to-back, then the MFMAs — one L2
+            // round-trip covers the batch instead of one per fragment.
+            char* rd_buf = smem + (t & 1) * tile_bytes;
             for (int kk = 0; kk < H; kk += F::KSTEP) {
                 int k = kk + a_kgrp;
                 typename F::T a_frag = *reinterpret_cast<const typename F::T*>(
-                    &smem[swz(a_row * rowbytes + k * sizeof(DT), a_row)]);
+                    &rd_buf[swz(a_row * rowbytes + k * sizeof(DT), a_row)]);
+                typename F::T breg[4][CHUNK];
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
 #pragma unroll
                     for (int mc = 0; mc < CHUNK; ++mc) {
                         int j = g * H + col_base + (cc * CHUNK + mc) * 16 + (lane & 15);
-                        typename F::T b_frag = *reinterpret_cast<const typename F::T*>(
+                        breg[g][mc] = *reinterpret_cast<const typename F::T*>(
                             &wh[(long)j * H + k]);
-                        acc[g][mc] = F::mfma(a_frag, b_frag, acc[g][mc]);
                     }
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int mc = 0; mc < CHUNK; ++mc)
+                        acc[g][mc] = F::mfma(a_frag, breg[g][mc], acc[g][mc]);
             }
             // ---- epilogue: activations, c update, global stores; h -> regs
 #pragma unroll
@@ -186,17 +205,18 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                 }
             }
         }
-        __syncthreads();  // all A-reads of h(t-1) done before h(t) LDS writes
+        char* wr_buf = smem + ((t + 1) & 1) * tile_bytes;
 #pragma unroll
         for (int mf = 0; mf < HFRAG; ++mf) {
             int col = col_base + mf * 16 + c_col;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = c_row0 + r;
-                st_lds<DT>(smem, swz(row * rowbytes + col * sizeof(DT), row), hn_reg[mf][r]);
+                st_lds<DT>(reinterpret_cast<char*>(wr_buf),
+                           swz(row * rowbytes + col * sizeof(DT), row), hn_reg[mf][r]);
             }
         }
-        __syncthreads();  // h(t) visible before next step's A-reads
+        __syncthreads();  // h(t) visible; h(t-1) buffer free for overwrite
     }
     // ---- final cell state
 #pragma unroll
@@ -211,7 +231,7 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
 
 // ---------------------------------------------------------------- backward
 template <typename DT, int HFRAG>
-__global__ void __launch_bounds__(512)
+__global__ void __launch_bounds__(HFRAG <= 2 ? 1024 : 512)
 lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
                 const DT* __restrict__ dhT,        // [B, H]
                 const DT* __restrict__ dcT,        // [B, H]
@@ -222,7 +242,7 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
                 DT* __restrict__ dgates,           // [T, B, 4H]
                 DT* __restrict__ dh0,              // [B, H]
                 DT* __restrict__ dc0,              // [B, H]
-                int T, int B, int H) {
+                int T, int B, int H, int db) {     // db: double-buffered LDS tile
     using F = Frag<DT>;
     extern __shared__ __attribute__((aligned(16))) char smem[];  // dgates tile [16][4H] DT, swizzled
 
@@ -260,6 +280,7 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
         const float* cs_t = c_seq + (long)t * B * H;
         const float* cs_p = (t > 0) ? c_seq + (long)(t - 1) * B * H : nullptr;
         DT* dg_t = dgates + (long)t * B * H4;
+        char* buf = smem + (db ? (t & 1) * (unsigned)(16 * rowbytes4) : 0u);
         // ---- lane-local dgates from stashed activations
 #pragma unroll
         for (int mf = 0; mf < HFRAG; ++mf) {
@@ -296,7 +317,7 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
                 // LDS stage for the recurrence MFMA (pad rows write zeros)
                 const float vals[4] = {dgi, dgf, dgg, dgo};
                 for (int g = 0; g < 4; ++g)
-                    st_lds<DT>(smem, swz(row * rowbytes4 + (g * H + col) * sizeof(DT), row),
+                    st_lds<DT>(buf, swz(row * rowbytes4 + (g * H + col) * sizeof(DT), row),
                                vals[g]);
             }
             // pin load movement at fragment boundaries: without this the
@@ -306,7 +327,7 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
         }
         __syncthreads();
         // ---- dh_prev = dgates @ Wh : MFMA over K = 4H (chunked, see fwd)
-        constexpr int CHUNK = (HFRAG > 4 && HFRAG % 4 == 0) ? 4 : HFRAG;
+        constexpr int CHUNK = (HFRAG > 2 && HFRAG % 2 == 0) ? 2 : HFRAG;
         constexpr int NCHUNK = HFRAG / CHUNK;
 #pragma unroll
         for (int cc = 0; cc < NCHUNK; ++cc) {
@@ -316,21 +337,24 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
             for (int kk = 0; kk < H4; kk += F::KSTEP) {
                 int k = kk + a_kgrp;
                 typename F::T a_frag = *reinterpret_cast<const typename F::T*>(
-                    &smem[swz(a_row * rowbytes4 + k * sizeof(DT), a_row)]);
+                    &buf[swz(a_row * rowbytes4 + k * sizeof(DT), a_row)]);
+                typename F::T breg[CHUNK];
 #pragma unroll
                 for (int mc = 0; mc < CHUNK; ++mc) {
                     int n = col_base + (cc * CHUNK + mc) * 16 + (lane & 15);
-                    typename F::T b_frag = *reinterpret_cast<const typename F::T*>(
+                    breg[mc] = *reinterpret_cast<const typename F::T*>(
                         &whT[(long)n * H4 + k]);
-                    acc[mc] = F::mfma(a_frag, b_frag, acc[mc]);
                 }
+#pragma unroll
+                for (int mc = 0; mc < CHUNK; ++mc)
+                    acc[mc] = F::mfma(a_frag, breg[mc], acc[mc]);
             }
 #pragma unroll
             for (int mc = 0; mc < CHUNK; ++mc)
 #pragma unroll
                 for (int r = 0; r < 4; ++r) dh_reg[cc * CHUNK + mc][r] = acc[mc][r];
         }
-        __syncthreads();  // LDS reused next iteration
+        if (!db) __syncthreads();  // single-buffer: tile reused next iteration
     }
     // ---- dh0 / dc0
 #pragma unroll
@@ -347,15 +371,28 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
 }
 
 // ------------------------------------------------------------- launchers
-// Wave-count selection: HFRAG (fragments per gate per wave) must stay <= 4
-// to avoid register spill, so H <= 256 runs 4-wave (256-thread) blocks and
-// H in {512, 1024} runs 8-wave (512-thread) blocks (2 waves/SIMD).
+// Wave-count selection. The kernel is L2-latency-bound on the per-step Wh
+// fragment loads, so prefer MANY SMALL waves (HFRAG=2 -> 4 waves/SIMD of
+// latency hiding) over few fat ones: waves = H/(16*HFRAG), block <= 1024
+// threads. LSTM_HF=1|2|4|8 in the env forces HFRAG for A/B experiments.
+#include <cstdlib>
 static inline int pick_geom(int H, int* threads, int* hf) {
     if (H % 64) return -1;
-    int f64 = H / 64;
-    if (f64 <= 4) { *threads = 256; *hf = f64; return 0; }
-    if (H % 128 == 0 && H / 128 <= 4) { *threads = 512; *hf = H / 128; return 0; }
-    if (f64 <= 8) { *threads = 256; *hf = f64; return 0; }  // odd large H: slow path
+    static int forced = [] { const char* e = getenv("LSTM_HF"); return e ? atoi(e) : 0; }();
+    const int prefs_forced[1] = {forced};
+    const int prefs_auto[4] = {2, 1, 4, 8};
+    const int* prefs = forced ? prefs_forced : prefs_auto;
+    int np = forced ? 1 : 4;
+    for (int i = 0; i < np; ++i) {
+        int f = prefs[i];
+        int cols = 16 * f;
+        if (H % cols) continue;
+        int waves = H / cols;
+        if (waves < 2 || waves > 16) continue;
+        *threads = waves * 64;
+        *hf = f;
+        return 0;
+    }
     return -1;
 }
 
@@ -378,7 +415,7 @@ extern "C" int launch_lstm_fwd_bf16(
     int T, int B, int H, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
-    size_t lds = (size_t)16 * H * sizeof(__hip_bfloat16);
+    size_t lds = (size_t)2 * 16 * H * sizeof(__hip_bfloat16);
     DISPATCH_HFRAG(hf,
         (lstm_fwd_kernel<__hip_bfloat16, HF><<<dim3(blocks), dim3(threads), lds, stream>>>(
             (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
@@ -394,7 +431,7 @@ extern "C" int launch_lstm_fwd_f32(
     int T, int B, int H, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
-    size_t lds = (size_t)16 * H * sizeof(float);
+    size_t lds = (size_t)2 * 16 * H * sizeof(float);
     DISPATCH_HFRAG(hf,
         (lstm_fwd_kernel<float, HF><<<dim3(blocks), dim3(threads), lds, stream>>>(
             (const float*)gates_x, (const float*)h0, (const float*)c0,
@@ -409,14 +446,16 @@ extern "C" int launch_lstm_bwd_bf16(
     void* dgates, void* dh0, void* dc0, int T, int B, int H, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
-    size_t lds = (size_t)16 * 4 * H * sizeof(__hip_bfloat16);
+    size_t tile = (size_t)16 * 4 * H * sizeof(__hip_bfloat16);
+    int db = (2 * tile <= 160 * 1024) ? 1 : 0;
+    size_t lds = tile * (db ? 2 : 1);
     DISPATCH_HFRAG(hf,
         (lstm_bwd_kernel<__hip_bfloat16, HF><<<dim3(blocks), dim3(threads), lds, stream>>>(
             (const __hip_bfloat16*)dh_seq, (const __hip_bfloat16*)dhT,
             (const __hip_bfloat16*)dcT, (const __hip_bfloat16*)gates_act, c_seq,
             (const __hip_bfloat16*)c0, (const __hip_bfloat16*)whT,
             (__hip_bfloat16*)dgates, (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0,
-            T, B, H)));
+            T, B, H, db)));
     return 0;
 }
 
@@ -426,11 +465,13 @@ extern "C" int launch_lstm_bwd_f32(
     void* dgates, void* dh0, void* dc0, int T, int B, int H, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
-    size_t lds = (size_t)16 * 4 * H * sizeof(float);
+    size_t tile = (size_t)16 * 4 * H * sizeof(float);
+    int db = (2 * tile <= 160 * 1024) ? 1 : 0;
+    size_t lds = tile * (db ? 2 : 1);
     DISPATCH_HFRAG(hf,
         (lstm_bwd_kernel<float, HF><<<dim3(blocks), dim3(threads), lds, stream>>>(
             (const float*)dh_seq, (const float*)dhT, (const float*)dcT,
             (const float*)gates_act, c_seq, (const float*)c0, (const float*)whT,
-            (float*)dgates, (float*)dh0, (float*)dc0, T, B, H)));
+            (float*)dgates, (float*)dh0, (float*)dc0, T, B, H, db)));
     return 0;
 }

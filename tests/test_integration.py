@@ -125,3 +125,62 @@ def test_cli_train_single_and_eval(tmp_path):
     assert rc == 0
     rc = main(["eval", "--config", str(cfgfile), "--episodes", "2"])
     assert rc == 0
+
+
+@pytest.mark.timeout(300)
+def test_pixel_learner_step_cpu():
+    """Pixel path: uint8 obs through Conv+LSTM trunks, normalized inside."""
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+
+    cfg = Config()
+    cfg.env.name = "quadruped-run-pixels"
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.model.conv_channels = (8, 8, 8)
+    cfg.replay.seq_len = 6
+    cfg.replay.burn_in = 2
+    cfg.replay.n_step = 2
+    cfg.learner.batch_size = 2
+    spec = make_env(cfg.env.name).spec
+    lrn = Learner(cfg, spec, device="cpu")
+    L = cfg.replay.seq_len + cfg.replay.n_step
+    batch = {
+        "obs": torch.randint(0, 255, (2, L, 84, 84, 3), dtype=torch.uint8),
+        "action": torch.randn(2, L, 12).clamp_(-1, 1),
+        "reward": torch.rand(2, L),
+        "done": torch.zeros(2, L),
+        "state": torch.zeros(2, 4, 16),
+        "weight": torch.ones(2),
+        "idx": None,
+    }
+    stats = lrn.train_step(batch)
+    assert np.isfinite(stats.critic_loss) and np.isfinite(stats.actor_loss)
+
+
+@pytest.mark.timeout(300)
+def test_pixel_actor_rollout_and_replay():
+    from pytorch_r2d2_dpg_amd.actor import ActorWorker
+    from pytorch_r2d2_dpg_amd.replay.gpu_buffer import GpuSequenceReplay
+
+    cfg = Config()
+    cfg.env.name = "quadruped-run-pixels"
+    cfg.env.time_limit = 30
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.model.conv_channels = (8, 8, 8)
+    cfg.replay.seq_len = 6
+    cfg.replay.burn_in = 2
+    cfg.replay.n_step = 2
+    cfg.replay.overlap = 3
+    w = ActorWorker(cfg, actor_id=0, sigma=0.3)
+    seqs = w.rollout(20)
+    assert seqs and seqs[0].obs.dtype == np.uint8
+    rep = GpuSequenceReplay(capacity=16, stored_len=8, obs_shape=(84, 84, 3),
+                            action_dim=12, hidden_size=16, device="cpu",
+                            obs_dtype=torch.uint8)
+    from pytorch_r2d2_dpg_amd.replay.staging import StagingRing
+    ring = StagingRing(rep, slots=2, slot_seqs=4)
+    ring.ingest(seqs)
+    b = rep.sample(4)
+    assert b["obs"].dtype == torch.uint8

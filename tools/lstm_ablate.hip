@@ -1,0 +1,180 @@
+// Standalone ablation of the fused-LSTM recurrent loop (tools/, not shipped).
+// Isolates which phase of lstm_fwd_kernel costs the time:
+//   V0: K-loop only (A from LDS, B from global, MFMA), h tile static
+//   V1: V0 + epilogue activations + LDS h writeback (no global I/O)
+//   V2: V1 + gates_x accumulator-init loads
+//   V3: full (V2 + h_seq/gates_act/c_seq global stores)
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/lstm_ablate.hip -o gpurun_out/lstm_ablate
+// Run:   ./gpurun_out/lstm_ablate [B] [H] [T]
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define DEV_INLINE __device__ __forceinline__
+
+DEV_INLINE float sigm(float x) { return 1.0f / (1.0f + __expf(-x)); }
+DEV_INLINE float tanh_e(float x) { return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f; }
+DEV_INLINE unsigned swz(unsigned b, int row) { return b ^ ((unsigned)(row & 7) << 4); }
+
+template <int HFRAG, int VARIANT>
+__global__ void __launch_bounds__(HFRAG <= 2 ? 1024 : 512)
+ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
+              const __hip_bfloat16* __restrict__ wh,
+              __hip_bfloat16* __restrict__ h_seq,
+              __hip_bfloat16* __restrict__ gates_act,
+              float* __restrict__ c_seq,
+              int T, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    const int brow0 = blockIdx.x * 16;
+    const int HQ = HFRAG * 16, col_base = wid * HQ;
+    const int H4 = 4 * H, rowbytes = H * 2;
+    const int a_row = lane & 15, a_kgrp = (lane >> 4) * 8;
+    const int c_col = lane & 15, c_row0 = (lane >> 4) * 4;
+
+    for (int e = tid; e < 16 * H; e += (int)blockDim.x) {
+        int r = e / H, col = e - r * H;
+        *reinterpret_cast<__hip_bfloat16*>(&smem[swz(r * rowbytes + col * 2, r)]) =
+            __float2bfloat16(0.01f * (float)(e & 7));
+    }
+    float c_reg[HFRAG][4];
+#pragma unroll
+    for (int mf = 0; mf < HFRAG; ++mf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) c_reg[mf][r] = 0.1f;
+    __syncthreads();
+
+    for (int t = 0; t < T; ++t) {
+        const __hip_bfloat16* gx_t = gates_x + (long)t * B * H4;
+        __hip_bfloat16* hs_t = h_seq + (long)t * B * H;
+        __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
+        float* cs_t = c_seq + (long)t * B * H;
+        f32x4 acc[4][HFRAG];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int mf = 0; mf < HFRAG; ++mf) {
+                if constexpr (VARIANT >= 2) {
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int grow = brow0 + c_row0 + r;
+                        int j = g * H + col_base + mf * 16 + c_col;
+                        acc[g][mf][r] = __bfloat162float(gx_t[(long)grow * H4 + j]);
+                    }
+                } else {
+                    acc[g][mf] = f32x4{0, 0, 0, 0};
+                }
+            }
+        for (int kk = 0; kk < H; kk += 32) {
+            int k = kk + a_kgrp;
+            short8 a_frag = *reinterpret_cast<const short8*>(
+                &smem[swz(a_row * rowbytes + k * 2, a_row)]);
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mf = 0; mf < HFRAG; ++mf) {
+                    int j = g * H + col_base + mf * 16 + (lane & 15);
+                    short8 b_frag = *reinterpret_cast<const short8*>(&wh[(long)j * H + k]);
+                    acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag, b_frag, acc[g][mf], 0, 0, 0);
+                }
+        }
+        __syncthreads();
+        if constexpr (VARIANT >= 1) {
+#pragma unroll
+            for (int mf = 0; mf < HFRAG; ++mf) {
+                int col = col_base + mf * 16 + c_col;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int row = c_row0 + r;
+                    int grow = brow0 + row;
+                    float gi = sigm(acc[0][mf][r]);
+                    float gf = sigm(acc[1][mf][r]);
+                    float gg = tanh_e(acc[2][mf][r]);
+                    float go = sigm(acc[3][mf][r]);
+                    float c = gf * c_reg[mf][r] + gi * gg;
+                    float hn = go * tanh_e(c);
+                    c_reg[mf][r] = c;
+                    *reinterpret_cast<__hip_bfloat16*>(
+                        &smem[swz(row * rowbytes + col * 2, row)]) = __float2bfloat16(hn);
+                    if constexpr (VARIANT >= 3) {
+                        if (grow < B) {
+                            hs_t[(long)grow * H + col] = __float2bfloat16(hn);
+                            cs_t[(long)grow * H + col] = c;
+                            ga_t[(long)grow * H4 + 0 * H + col] = __float2bfloat16(gi);
+                            ga_t[(long)grow * H4 + 1 * H + col] = __float2bfloat16(gf);
+                            ga_t[(long)grow * H4 + 2 * H + col] = __float2bfloat16(gg);
+                            ga_t[(long)grow * H4 + 3 * H + col] = __float2bfloat16(go);
+                        }
+                    }
+                }
+            }
+        } else {
+            // keep acc live without the epilogue (DCE guard, guide rule 17)
+#pragma unroll
+            for (int mf = 0; mf < HFRAG; ++mf) {
+                asm volatile("" ::"v"(acc[0][mf][0]));
+                if (acc[0][mf][0] == 1234.5f)
+                    *reinterpret_cast<__hip_bfloat16*>(&smem[swz(c_col * 2, 0)]) =
+                        __float2bfloat16(acc[0][mf][0]);
+            }
+        }
+        __syncthreads();
+    }
+}
+
+#define CHECK(x) do { hipError_t e = (x); if (e) { printf("ERR %s @%d\n", hipGetErrorString(e), __LINE__); exit(1); } } while (0)
+
+template <int HFRAG, int VARIANT>
+float run_variant(int B, int H, int T, int iters) {
+    long n_gx = (long)T * B * 4 * H, n_h = (long)T * B * H;
+    __hip_bfloat16 *gx, *wh, *hs, *ga;
+    float* cs;
+    CHECK(hipMalloc(&gx, n_gx * 2));
+    CHECK(hipMalloc(&wh, (long)4 * H * H * 2));
+    CHECK(hipMalloc(&hs, n_h * 2));
+    CHECK(hipMalloc(&ga, n_gx * 2));
+    CHECK(hipMalloc(&cs, n_h * 4));
+    CHECK(hipMemset(gx, 0x3c, n_gx * 2));
+    CHECK(hipMemset(wh, 0x3a, (long)4 * H * H * 2));
+    int blocks = (B + 15) / 16;
+    int waves = H / (16 * HFRAG);
+    dim3 thr(waves * 64);
+    size_t lds = (size_t)16 * H * 2;
+    hipEvent_t e0, e1;
+    CHECK(hipEventCreate(&e0));
+    CHECK(hipEventCreate(&e1));
+    for (int i = 0; i < 3; ++i)
+        ablate_kernel<HFRAG, VARIANT><<<blocks, thr, lds>>>(gx, wh, hs, ga, cs, T, B, H);
+    CHECK(hipDeviceSynchronize());
+    CHECK(hipEventRecord(e0));
+    for (int i = 0; i < iters; ++i)
+        ablate_kernel<HFRAG, VARIANT><<<blocks, thr, lds>>>(gx, wh, hs, ga, cs, T, B, H);
+    CHECK(hipEventRecord(e1));
+    CHECK(hipEventSynchronize(e1));
+    float ms;
+    CHECK(hipEventElapsedTime(&ms, e0, e1));
+    hipFree(gx); hipFree(wh); hipFree(hs); hipFree(ga); hipFree(cs);
+    return ms / iters;
+}
+
+int main(int argc, char** argv) {
+    int B = argc > 1 ? atoi(argv[1]) : 256;
+    int H = argc > 2 ? atoi(argv[2]) : 512;
+    int T = argc > 3 ? atoi(argv[3]) : 40;
+    if (H != 512) { printf("H=512 only in this probe\n"); return 1; }
+    printf("B=%d H=%d T=%d  (16 blocks x %d waves, HFRAG=2)\n", B, H, T, H / 32);
+    printf("V0 K-loop only          : %8.3f ms\n", run_variant<2, 0>(B, H, T, 10));
+    printf("V1 +epilogue+LDS-h      : %8.3f ms\n", run_variant<2, 1>(B, H, T, 10));
+    printf("V2 +gates_x acc init    : %8.3f ms\n", run_variant<2, 2>(B, H, T, 10));
+    printf("V3 full (global stores) : %8.3f ms\n", run_variant<2, 3>(B, H, T, 10));
+    // HFRAG=4 (8-wave) full for comparison
+    printf("V3 full HFRAG=4 8-wave  : %8.3f ms\n", run_variant<4, 3>(B, H, T, 10));
+    return 0;
+}

@@ -39,6 +39,19 @@ hipStream_t cur_stream() {
     return at::hip::getCurrentHIPStream().stream();
 }
 
+// Pack a [R, C] weight into the MFMA-fragment-major layout the LSTM kernels
+// load with lane-linear coalesced instructions: [R/16][C/KSTEP][64][KPL],
+// where lane l of block (rb, cb) holds row rb*16+(l&15), cols
+// cb*KSTEP+(l>>4)*KPL ... +KPL. KSTEP/KPL: bf16 32/8, f32 4/1.
+torch::Tensor pack_fragment_major(const torch::Tensor& w) {
+    const auto R = w.size(0), C = w.size(1);
+    const int kstep = w.scalar_type() == torch::kBFloat16 ? 32 : 4;
+    const int kpl = kstep / 4;
+    return w.view({R / 16, 16, C / kstep, 4, kpl})
+        .permute({0, 2, 3, 1, 4})
+        .contiguous();
+}
+
 void check_dev_contig(const torch::Tensor& t, const char* name) {
     TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
     TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
@@ -63,6 +76,7 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
     TORCH_CHECK(h0.scalar_type() == dt && wh.scalar_type() == dt && c0.scalar_type() == dt,
                 "dtype mismatch");
 
+    auto wh_p = pack_fragment_major(wh);
     auto h_seq = torch::empty({T, B, H}, gates_x.options());
     auto c_final = torch::empty({B, H}, gates_x.options());
     auto gates_act = torch::empty({T, B, 4 * H}, gates_x.options());
@@ -71,12 +85,12 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
     int rc;
     if (dt == torch::kBFloat16)
         rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
-                                  wh.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
+                                  wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
                                   gates_act.data_ptr(), c_seq.data_ptr<float>(),
                                   (int)T, (int)B, (int)H, cur_stream());
     else
         rc = launch_lstm_fwd_f32(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
-                                 wh.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
+                                 wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
                                  (int)T, (int)B, (int)H, cur_stream());
     TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);
@@ -101,6 +115,7 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
     TORCH_CHECK(whT.size(0) == H && whT.size(1) == 4 * H, "whT must be [H, 4H]");
     auto dt = dh_seq.scalar_type();
 
+    auto whT_p = pack_fragment_major(whT);
     auto dgates = torch::empty({T, B, 4 * H}, dh_seq.options());
     auto dh0 = torch::empty({B, H}, dh_seq.options());
     auto dc0 = torch::empty({B, H}, dh_seq.options());
@@ -109,13 +124,13 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
     if (dt == torch::kBFloat16)
         rc = launch_lstm_bwd_bf16(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
                                   gates_act.data_ptr(), c_seq.data_ptr<float>(),
-                                  c0.data_ptr(), whT.data_ptr(), dgates.data_ptr(),
+                                  c0.data_ptr(), whT_p.data_ptr(), dgates.data_ptr(),
                                   dh0.data_ptr(), dc0.data_ptr(),
                                   (int)T, (int)B, (int)H, cur_stream());
     else
         rc = launch_lstm_bwd_f32(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
-                                 c0.data_ptr(), whT.data_ptr(), dgates.data_ptr(),
+                                 c0.data_ptr(), whT_p.data_ptr(), dgates.data_ptr(),
                                  dh0.data_ptr(), dc0.data_ptr(),
                                  (int)T, (int)B, (int)H, cur_stream());
     TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);

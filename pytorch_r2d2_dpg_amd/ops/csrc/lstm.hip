@@ -164,13 +164,18 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                 typename F::T a_frag = *reinterpret_cast<const typename F::T*>(
                     &rd_buf[swz(a_row * rowbytes + k * sizeof(DT), a_row)]);
                 typename F::T breg[4][CHUNK];
+                // wh is packed fragment-major ([4H/16][H/KSTEP][64][KPL]):
+                // every load is one lane-linear, fully-coalesced 16B/lane
+                // instruction (unpacked row-major cost 16 scattered lines
+                // per instruction -- measured as 86% of kernel time)
+                const int kb = kk / F::KSTEP;
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
 #pragma unroll
                     for (int mc = 0; mc < CHUNK; ++mc) {
-                        int j = g * H + col_base + (cc * CHUNK + mc) * 16 + (lane & 15);
-                        breg[g][mc] = *reinterpret_cast<const typename F::T*>(
-                            &wh[(long)j * H + k]);
+                        int jb = g * (H >> 4) + (col_base >> 4) + cc * CHUNK + mc;
+                        long off = (((long)jb * (H / F::KSTEP) + kb) * 64 + lane) * F::KPL;
+                        breg[g][mc] = *reinterpret_cast<const typename F::T*>(&wh[off]);
                     }
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
@@ -339,11 +344,12 @@ lstm_bwd_kernel(const DT* __restrict__ dh_seq,     // [T, B, H]
                 typename F::T a_frag = *reinterpret_cast<const typename F::T*>(
                     &buf[swz(a_row * rowbytes4 + k * sizeof(DT), a_row)]);
                 typename F::T breg[CHUNK];
+                const int kb = kk / F::KSTEP;
 #pragma unroll
                 for (int mc = 0; mc < CHUNK; ++mc) {
-                    int n = col_base + (cc * CHUNK + mc) * 16 + (lane & 15);
-                    breg[mc] = *reinterpret_cast<const typename F::T*>(
-                        &whT[(long)n * H4 + k]);
+                    int nb = (col_base >> 4) + cc * CHUNK + mc;
+                    long off = (((long)nb * (H4 / F::KSTEP) + kb) * 64 + lane) * F::KPL;
+                    breg[mc] = *reinterpret_cast<const typename F::T*>(&whT[off]);
                 }
 #pragma unroll
                 for (int mc = 0; mc < CHUNK; ++mc)

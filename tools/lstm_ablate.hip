@@ -22,6 +22,9 @@ DEV_INLINE float sigm(float x) { return 1.0f / (1.0f + __expf(-x)); }
 DEV_INLINE float tanh_e(float x) { return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f; }
 DEV_INLINE unsigned swz(unsigned b, int row) { return b ^ ((unsigned)(row & 7) << 4); }
 
+// VARIANT: low 2 bits = build level (0..3); bit3 skip barriers; bit4 fixed
+// B address (always same 16B -> no L2 pressure); bit5 constant A frag (no
+// LDS read); bit6 simple epilogue (no transcendentals)
 template <int HFRAG, int VARIANT>
 __global__ void __launch_bounds__(HFRAG <= 2 ? 1024 : 512)
 ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
@@ -60,7 +63,7 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
         for (int g = 0; g < 4; ++g)
 #pragma unroll
             for (int mf = 0; mf < HFRAG; ++mf) {
-                if constexpr (VARIANT >= 2) {
+                if constexpr ((VARIANT & 3) >= 2) {
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
                         int grow = brow0 + c_row0 + r;
@@ -73,20 +76,30 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
             }
         for (int kk = 0; kk < H; kk += 32) {
             int k = kk + a_kgrp;
-            short8 a_frag = *reinterpret_cast<const short8*>(
-                &smem[swz(a_row * rowbytes + k * 2, a_row)]);
+            short8 a_frag;
+            if constexpr ((VARIANT & 32) == 0)
+                a_frag = *reinterpret_cast<const short8*>(
+                    &smem[swz(a_row * rowbytes + k * 2, a_row)]);
+            else
+                a_frag = short8{1, 2, 3, 4, 5, 6, 7, 8};
 #pragma unroll
             for (int g = 0; g < 4; ++g)
 #pragma unroll
                 for (int mf = 0; mf < HFRAG; ++mf) {
-                    int j = g * H + col_base + mf * 16 + (lane & 15);
-                    short8 b_frag = *reinterpret_cast<const short8*>(&wh[(long)j * H + k]);
+                    long boff;
+                    if constexpr ((VARIANT & 16) == 0) {
+                        int j = g * H + col_base + mf * 16 + (lane & 15);
+                        boff = (long)j * H + k;
+                    } else {
+                        boff = (lane & 15) * 8;  // fixed hot line set
+                    }
+                    short8 b_frag = *reinterpret_cast<const short8*>(&wh[boff]);
                     acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag, b_frag, acc[g][mf], 0, 0, 0);
                 }
         }
-        __syncthreads();
-        if constexpr (VARIANT >= 1) {
+        if constexpr ((VARIANT & 8) == 0) __syncthreads();
+        if constexpr ((VARIANT & 3) >= 1) {
 #pragma unroll
             for (int mf = 0; mf < HFRAG; ++mf) {
                 int col = col_base + mf * 16 + c_col;
@@ -94,16 +107,24 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
                 for (int r = 0; r < 4; ++r) {
                     int row = c_row0 + r;
                     int grow = brow0 + row;
-                    float gi = sigm(acc[0][mf][r]);
-                    float gf = sigm(acc[1][mf][r]);
-                    float gg = tanh_e(acc[2][mf][r]);
-                    float go = sigm(acc[3][mf][r]);
-                    float c = gf * c_reg[mf][r] + gi * gg;
-                    float hn = go * tanh_e(c);
+                    float gi, gf, gg, go, c, hn;
+                    if constexpr ((VARIANT & 64) == 0) {
+                        gi = sigm(acc[0][mf][r]);
+                        gf = sigm(acc[1][mf][r]);
+                        gg = tanh_e(acc[2][mf][r]);
+                        go = sigm(acc[3][mf][r]);
+                        c = gf * c_reg[mf][r] + gi * gg;
+                        hn = go * tanh_e(c);
+                    } else {
+                        gi = acc[0][mf][r]; gf = acc[1][mf][r];
+                        gg = acc[2][mf][r]; go = acc[3][mf][r];
+                        c = gf * c_reg[mf][r] + gi * gg;
+                        hn = go * 0.5f + c * 0.25f;
+                    }
                     c_reg[mf][r] = c;
                     *reinterpret_cast<__hip_bfloat16*>(
                         &smem[swz(row * rowbytes + col * 2, row)]) = __float2bfloat16(hn);
-                    if constexpr (VARIANT >= 3) {
+                    if constexpr ((VARIANT & 3) >= 3) {
                         if (grow < B) {
                             hs_t[(long)grow * H + col] = __float2bfloat16(hn);
                             cs_t[(long)grow * H + col] = c;
@@ -170,11 +191,14 @@ int main(int argc, char** argv) {
     int T = argc > 3 ? atoi(argv[3]) : 40;
     if (H != 512) { printf("H=512 only in this probe\n"); return 1; }
     printf("B=%d H=%d T=%d  (16 blocks x %d waves, HFRAG=2)\n", B, H, T, H / 32);
-    printf("V0 K-loop only          : %8.3f ms\n", run_variant<2, 0>(B, H, T, 10));
-    printf("V1 +epilogue+LDS-h      : %8.3f ms\n", run_variant<2, 1>(B, H, T, 10));
-    printf("V2 +gates_x acc init    : %8.3f ms\n", run_variant<2, 2>(B, H, T, 10));
-    printf("V3 full (global stores) : %8.3f ms\n", run_variant<2, 3>(B, H, T, 10));
-    // HFRAG=4 (8-wave) full for comparison
-    printf("V3 full HFRAG=4 8-wave  : %8.3f ms\n", run_variant<4, 3>(B, H, T, 10));
+    printf("V0  K-loop only           : %8.3f ms\n", run_variant<2, 0>(B, H, T, 10));
+    printf("V1  +epilogue+LDS-h       : %8.3f ms\n", run_variant<2, 1>(B, H, T, 10));
+    printf("V3  full                  : %8.3f ms\n", run_variant<2, 3>(B, H, T, 10));
+    printf("V1n V1 no barriers        : %8.3f ms\n", run_variant<2, 1 | 8>(B, H, T, 10));
+    printf("V1b V1 fixed-B-addr       : %8.3f ms\n", run_variant<2, 1 | 16>(B, H, T, 10));
+    printf("V1a V1 const A (no LDSrd) : %8.3f ms\n", run_variant<2, 1 | 32>(B, H, T, 10));
+    printf("V1e V1 simple epilogue    : %8.3f ms\n", run_variant<2, 1 | 64>(B, H, T, 10));
+    printf("V1x V1 fixedB+constA+sepi : %8.3f ms\n", run_variant<2, 1 | 16 | 32 | 64>(B, H, T, 10));
+    printf("V1y V1 nobar+fixedB       : %8.3f ms\n", run_variant<2, 1 | 8 | 16>(B, H, T, 10));
     return 0;
 }

@@ -139,7 +139,7 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
         DT* hs_t = h_seq + (long)t * B * H;
         DT* ga_t = gates_act + (long)t * B * H4;
         float* cs_t = c_seq + (long)t * B * H;
-        float hn_reg[HFRAG][4];
+        char* wr_buf = smem + ((t + 1) & 1) * tile_bytes;
 #pragma unroll
         for (int cc = 0; cc < NCHUNK; ++cc) {
             // accumulators initialized with gates_x (bias + input projection)
@@ -183,14 +183,16 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                     for (int mc = 0; mc < CHUNK; ++mc)
                         acc[g][mc] = F::mfma(a_frag, breg[g][mc], acc[g][mc]);
             }
-            // ---- epilogue: activations, c update, global stores; h -> regs
+            // ---- epilogue: activations, c update, LDS write into the OTHER
+            // buffer (no hazard: readers use rd_buf) + global stores
 #pragma unroll
             for (int mc = 0; mc < CHUNK; ++mc) {
                 int mf = cc * CHUNK + mc;
                 int col = col_base + mf * 16 + c_col;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    int grow = brow0 + c_row0 + r;
+                    int row = c_row0 + r;
+                    int grow = brow0 + row;
                     float gi = sigm(acc[0][mc][r]);
                     float gf = sigm(acc[1][mc][r]);
                     float gg = tanh_e(acc[2][mc][r]);
@@ -198,7 +200,7 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                     float c = gf * c_reg[mf][r] + gi * gg;
                     float hn = go * tanh_e(c);
                     c_reg[mf][r] = c;
-                    hn_reg[mf][r] = hn;
+                    st_lds<DT>(wr_buf, swz(row * rowbytes + col * sizeof(DT), row), hn);
                     if (grow < B) {
                         st_f(&hs_t[(long)grow * H + col], hn);
                         cs_t[(long)grow * H + col] = c;
@@ -208,17 +210,6 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                         st_f(&ga_t[(long)grow * H4 + 3 * H + col], go);
                     }
                 }
-            }
-        }
-        char* wr_buf = smem + ((t + 1) & 1) * tile_bytes;
-#pragma unroll
-        for (int mf = 0; mf < HFRAG; ++mf) {
-            int col = col_base + mf * 16 + c_col;
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int row = c_row0 + r;
-                st_lds<DT>(reinterpret_cast<char*>(wr_buf),
-                           swz(row * rowbytes + col * sizeof(DT), row), hn_reg[mf][r]);
             }
         }
         __syncthreads();  // h(t) visible; h(t-1) buffer free for overwrite

@@ -18,8 +18,12 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define DEV_INLINE __device__ __forceinline__
 
-DEV_INLINE float sigm(float x) { return 1.0f / (1.0f + __expf(-x)); }
-DEV_INLINE float tanh_e(float x) { return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f; }
+DEV_INLINE float sigm(float x) {
+    return __builtin_amdgcn_rcpf(1.0f + __builtin_amdgcn_exp2f(-1.44269504f * x));
+}
+DEV_INLINE float tanh_e(float x) {
+    return 2.0f * __builtin_amdgcn_rcpf(1.0f + __builtin_amdgcn_exp2f(-2.88539008f * x)) - 1.0f;
+}
 DEV_INLINE unsigned swz(unsigned b, int row) { return b ^ ((unsigned)(row & 7) << 4); }
 
 // VARIANT: low 2 bits = build level (0..3); bit3 skip barriers; bit4 fixed
@@ -35,6 +39,7 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
               int T, int B, int H) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    const unsigned tile_bytes = 16u * (unsigned)(H * 2);
     const int brow0 = blockIdx.x * 16;
     const int HQ = HFRAG * 16, col_base = wid * HQ;
     const int H4 = 4 * H, rowbytes = H * 2;
@@ -74,12 +79,13 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
                     acc[g][mf] = f32x4{0, 0, 0, 0};
                 }
             }
+        char* rd_buf = smem + (t & 1) * tile_bytes;
         for (int kk = 0; kk < H; kk += 32) {
             int k = kk + a_kgrp;
             short8 a_frag;
             if constexpr ((VARIANT & 32) == 0)
                 a_frag = *reinterpret_cast<const short8*>(
-                    &smem[swz(a_row * rowbytes + k * 2, a_row)]);
+                    &rd_buf[swz(a_row * rowbytes + k * 2, a_row)]);
             else
                 a_frag = short8{1, 2, 3, 4, 5, 6, 7, 8};
 #pragma unroll
@@ -88,17 +94,16 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
                 for (int mf = 0; mf < HFRAG; ++mf) {
                     long boff;
                     if constexpr ((VARIANT & 16) == 0) {
-                        int j = g * H + col_base + mf * 16 + (lane & 15);
-                        boff = (long)j * H + k;
+                        int jb = g * (H >> 4) + (col_base >> 4) + mf;
+                        boff = (((long)jb * (H >> 5) + (kk >> 5)) * 64 + lane) * 8;
                     } else {
-                        boff = (lane & 15) * 8;  // fixed hot line set
+                        boff = (long)lane * 8;  // fixed hot line set
                     }
                     short8 b_frag = *reinterpret_cast<const short8*>(&wh[boff]);
                     acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag, b_frag, acc[g][mf], 0, 0, 0);
                 }
         }
-        if constexpr ((VARIANT & 8) == 0) __syncthreads();
         if constexpr ((VARIANT & 3) >= 1) {
 #pragma unroll
             for (int mf = 0; mf < HFRAG; ++mf) {
@@ -123,7 +128,8 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
                     }
                     c_reg[mf][r] = c;
                     *reinterpret_cast<__hip_bfloat16*>(
-                        &smem[swz(row * rowbytes + col * 2, row)]) = __float2bfloat16(hn);
+                        &smem[(unsigned)((~t & 1)) * tile_bytes +
+                              swz(row * rowbytes + col * 2, row)]) = __float2bfloat16(hn);
                     if constexpr ((VARIANT & 3) >= 3) {
                         if (grow < B) {
                             hs_t[(long)grow * H + col] = __float2bfloat16(hn);
@@ -146,7 +152,7 @@ ablate_kernel(const __hip_bfloat16* __restrict__ gates_x,
                         __float2bfloat16(acc[0][mf][0]);
             }
         }
-        __syncthreads();
+        if constexpr ((VARIANT & 8) == 0) __syncthreads();
     }
 }
 
@@ -167,7 +173,7 @@ float run_variant(int B, int H, int T, int iters) {
     int blocks = (B + 15) / 16;
     int waves = H / (16 * HFRAG);
     dim3 thr(waves * 64);
-    size_t lds = (size_t)16 * H * 2;
+    size_t lds = (size_t)2 * 16 * H * 2;
     hipEvent_t e0, e1;
     CHECK(hipEventCreate(&e0));
     CHECK(hipEventCreate(&e1));

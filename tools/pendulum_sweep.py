@@ -40,11 +40,11 @@ def run_variant(args):
 if __name__ == "__main__":
     steps = int(sys.argv[1]) if len(sys.argv) > 1 else 1500
     grid = list(itertools.product(
-        [1.0, 0.1],        # reward_scale
-        [0.3, 0.8],        # sigma_max
-        [0.0, 1e-2],       # critic_weight_decay
+        [0.05, 0.1],       # reward_scale
+        [0.3, 0.5],        # sigma_max
+        [0.0],             # critic_weight_decay
         [3e-4],            # lr_actor
-        [steps], [0],
+        [steps], [0, 1],
     ))
     with mp.get_context("spawn").Pool(8) as pool:
         for args, ret in pool.imap_unordered(run_variant, grid):

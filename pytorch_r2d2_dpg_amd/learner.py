@@ -21,6 +21,7 @@ hundreds of launches into one.
 
 from __future__ import annotations
 
+import contextlib
 import time
 from dataclasses import dataclass
 from typing import Dict, Optional
@@ -35,6 +36,10 @@ from .ops.td import td_targets_and_priorities
 from .ops.reference import value_rescale, value_rescale_inv
 from .parallel.dist import DistContext
 from .replay.buffer import SequenceReplay
+
+
+def _null_ctx():
+    return contextlib.nullcontext()
 
 
 @dataclass
@@ -90,15 +95,37 @@ class Learner:
         self._static: Optional[Dict[str, torch.Tensor]] = None
         self._graph_stats: Optional[Dict[str, torch.Tensor]] = None
 
+        # stream-overlap machinery (GPU only): the actor chain runs on a side
+        # stream concurrently with the target/critic chain
+        self._actor_params = list(self.actor.parameters())
+        if self.device.type == "cuda":
+            self._side_stream = torch.cuda.Stream()
+            self._ev_ba = torch.cuda.Event()
+            self._ev_bc = torch.cuda.Event()
+            self._ev_cf = torch.cuda.Event()
+            self._ev_ab = torch.cuda.Event()
+        else:
+            self._side_stream = None
+            self._ev_ba = self._ev_bc = self._ev_cf = self._ev_ab = None
+
     # ---------------------------------------------------------- core update
     def _step_core(self, batch: Dict[str, torch.Tensor],
                    update_priorities: bool = True,
                    zero_to_none: bool = True) -> Dict[str, torch.Tensor]:
-        """The full device-side update. No host syncs, no Python floats."""
+        """The full device-side update. No host syncs, no Python floats.
+
+        The actor update evaluates Q through the PRE-update critic (standard
+        DDPG variant), which makes the actor chain independent of the critic
+        optimizer — on GPU the two chains run on separate HIP streams (the
+        LSTM unroll kernels occupy ~16 CUs each, so overlapping independent
+        unrolls is nearly free parallelism). The CPU path runs the identical
+        math on one stream.
+        """
         cfg = self.cfg
         Tn = self.train_len + self.n_step
         bi = self.burn_in
         T = self.train_len
+        overlap = self.device.type == "cuda" and self._side_stream is not None
 
         obs = batch["obs"].transpose(0, 1)                         # [L, B, ...]
         if obs.dtype != torch.uint8:  # pixel obs stay uint8 (trunk normalizes)
@@ -111,24 +138,57 @@ class Learner:
         h_a, c_a = st[:, 0].contiguous(), st[:, 1].contiguous()
         h_c, c_c = st[:, 2].contiguous(), st[:, 3].contiguous()
 
-        # ---- burn-in (no grad): rebuild states at the training-window start
+        s1 = self._side_stream
+        ev_ba = self._ev_ba
+        ev_bc = self._ev_bc
+        ev_cf = self._ev_cf
+        ev_ab = self._ev_ab
+
+        main = torch.cuda.current_stream() if overlap else None
+        if overlap:
+            s1.wait_stream(main)
+
+        # ---- burn-ins: actor chain on main, critic chain on the side stream
         if bi > 0:
             with torch.no_grad():
                 _, (h_a2, c_a2) = self.actor(obs[:bi], (h_a, c_a))
-                _, (h_c2, c_c2) = self.critic(obs[:bi], action[:bi], (h_c, c_c))
         else:
-            h_a2, c_a2, h_c2, c_c2 = h_a, c_a, h_c, c_c
+            h_a2, c_a2 = h_a, c_a
+        if overlap:
+            ev_ba.record(main)
+        with torch.cuda.stream(s1) if overlap else _null_ctx():
+            if bi > 0:
+                with torch.no_grad():
+                    _, (h_c2, c_c2) = self.critic(obs[:bi], action[:bi], (h_c, c_c))
+            else:
+                h_c2, c_c2 = h_c, c_c
+            if overlap:
+                ev_bc.record(s1)
+            # critic online fwd (needs only the critic burn-in state)
+            q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
+                                      (h_c2.detach(), c_c2.detach()))
+            q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
+            if overlap:
+                ev_cf.record(s1)
+                s1.wait_event(ev_ba)
+            # actor chain through the PRE-update critic
+            mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
+            q_pi, _ = self.critic(obs[bi:bi + T], mu, (h_c2.detach(), c_c2.detach()))
+            actor_loss = -q_pi.float().mean()
+            actor_grads = torch.autograd.grad(actor_loss, self._actor_params)
+            if overlap:
+                ev_ab.record(s1)
 
-        # ---- target unroll (no grad) over train window + n-step tail
+        # ---- target unroll (no grad) on main
         with torch.no_grad():
             mu_t, _ = self.target_actor(obs[bi:bi + Tn], (h_a2, c_a2))
+            if overlap:
+                main.wait_event(ev_bc)
             qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
             q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
 
-        # ---- critic update on n-step targets
-        q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
-                                  (h_c2.detach(), c_c2.detach()))
-        q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
+        if overlap:
+            main.wait_event(ev_cf)
         with torch.no_grad():
             y, td_abs, prio = td_targets_and_priorities(
                 reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),
@@ -141,20 +201,21 @@ class Learner:
         self.dist.allreduce_grads(self.critic.parameters(), key="critic")
         if cfg.learner.max_grad_norm > 0:
             torch.nn.utils.clip_grad_norm_(self.critic.parameters(), cfg.learner.max_grad_norm)
+        # the actor chain reads critic params: join before the optimizer writes
+        if overlap:
+            main.wait_event(ev_ab)
         self.opt_critic.step()
 
-        # ---- actor update: ascend Q(s, mu(s)) through the critic
-        mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
-        q_pi, _ = self.critic(obs[bi:bi + T], mu, (h_c2.detach(), c_c2.detach()))
-        actor_loss = -q_pi.float().mean()
-        self.opt_actor.zero_grad(set_to_none=zero_to_none)
-        self.opt_critic.zero_grad(set_to_none=zero_to_none)  # discard this path's grads
-        actor_loss.backward()
-        self.dist.allreduce_grads(self.actor.parameters(), key="actor")
+        # ---- actor optimizer (grads computed on the side stream)
+        for p, g in zip(self._actor_params, actor_grads):
+            if p.grad is None:
+                p.grad = g.clone() if not zero_to_none else g
+            else:
+                p.grad.copy_(g)
+        self.dist.allreduce_grads(self._actor_params, key="actor")
         if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(self.actor.parameters(), cfg.learner.max_grad_norm)
+            torch.nn.utils.clip_grad_norm_(self._actor_params, cfg.learner.max_grad_norm)
         self.opt_actor.step()
-        self.opt_critic.zero_grad(set_to_none=zero_to_none)
 
         # ---- Polyak soft target update (K4: one fused launch per net on GPU)
         # (branch is Python-side: under graph capture target_update_every == 1)

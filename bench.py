@@ -84,8 +84,9 @@ def main() -> int:
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--replay-capacity", type=int, default=8192)
     p.add_argument("--device", type=str, default="")
-    p.add_argument("--graph", action="store_true", default=True,
-                   help="capture the learner step in a hipGraph (default on)")
+    p.add_argument("--graph", action="store_true", default=False,
+                   help="capture the learner step in a hipGraph (default off: "
+                        "capture serializes the multi-stream overlap)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
     args = p.parse_args()
 

@@ -164,12 +164,6 @@ class Learner:
                 h_c2, c_c2 = h_c, c_c
             if overlap:
                 ev_bc.record(s1)
-            # critic online fwd (needs only the critic burn-in state)
-            q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
-                                      (h_c2.detach(), c_c2.detach()))
-            q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
-            if overlap:
-                ev_cf.record(s1)
                 s1.wait_event(ev_ba)
             # actor chain through the PRE-update critic
             mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
@@ -179,16 +173,17 @@ class Learner:
             if overlap:
                 ev_ab.record(s1)
 
-        # ---- target unroll (no grad) on main
+        # ---- target unroll + critic online fwd (main stream; balances the
+        # side stream's actor-chain length)
         with torch.no_grad():
             mu_t, _ = self.target_actor(obs[bi:bi + Tn], (h_a2, c_a2))
             if overlap:
                 main.wait_event(ev_bc)
             qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
             q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
-
-        if overlap:
-            main.wait_event(ev_cf)
+        q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
+                                  (h_c2.detach(), c_c2.detach()))
+        q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
         with torch.no_grad():
             y, td_abs, prio = td_targets_and_priorities(
                 reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),

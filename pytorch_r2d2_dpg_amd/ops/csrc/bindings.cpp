@@ -11,9 +11,11 @@
 
 // ---- extern "C" launchers from the .hip TUs ------------------------------
 extern "C" int launch_lstm_fwd_bf16(const void*, const void*, const void*, const void*,
-                                    void*, void*, void*, float*, int, int, int, hipStream_t);
+                                    void*, void*, void*, float*, int, int, int, int,
+                                    hipStream_t);
 extern "C" int launch_lstm_fwd_f32(const void*, const void*, const void*, const void*,
-                                   void*, void*, void*, float*, int, int, int, hipStream_t);
+                                   void*, void*, void*, float*, int, int, int, int,
+                                   hipStream_t);
 extern "C" int launch_lstm_bwd_bf16(const void*, const void*, const void*, const void*,
                                     const float*, const void*, const void*, void*, void*,
                                     void*, int, int, int, hipStream_t);
@@ -60,7 +62,7 @@ void check_dev_contig(const torch::Tensor& t, const char* name) {
 // ---- K1: fused LSTM sequence forward -------------------------------------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
-             torch::Tensor wh) {
+             torch::Tensor wh, bool stash = true) {
     check_dev_contig(gates_x, "gates_x");
     check_dev_contig(h0, "h0");
     check_dev_contig(c0, "c0");
@@ -79,20 +81,23 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
     auto wh_p = pack_fragment_major(wh);
     auto h_seq = torch::empty({T, B, H}, gates_x.options());
     auto c_final = torch::empty({B, H}, gates_x.options());
-    auto gates_act = torch::empty({T, B, 4 * H}, gates_x.options());
-    auto c_seq = torch::empty({T, B, H}, gates_x.options().dtype(torch::kFloat32));
+    auto gates_act = torch::empty({stash ? T : 0, B, 4 * H}, gates_x.options());
+    auto c_seq = torch::empty({stash ? T : 0, B, H},
+                              gates_x.options().dtype(torch::kFloat32));
 
     int rc;
     if (dt == torch::kBFloat16)
         rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
                                   wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
-                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
-                                  (int)T, (int)B, (int)H, cur_stream());
+                                  gates_act.data_ptr(),
+                                  stash ? c_seq.data_ptr<float>() : nullptr,
+                                  (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
     else
         rc = launch_lstm_fwd_f32(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
                                  wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
-                                 gates_act.data_ptr(), c_seq.data_ptr<float>(),
-                                 (int)T, (int)B, (int)H, cur_stream());
+                                 gates_act.data_ptr(),
+                                 stash ? c_seq.data_ptr<float>() : nullptr,
+                                 (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
     TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);
     return {h_seq, c_final, gates_act, c_seq};
 }
@@ -235,7 +240,9 @@ torch::Tensor fp8_gather_decode(torch::Tensor store, torch::Tensor idx,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("lstm_seq_fwd", &lstm_seq_fwd, "fused LSTM sequence forward (K1)");
+    m.def("lstm_seq_fwd", &lstm_seq_fwd, "fused LSTM sequence forward (K1)",
+          py::arg("gates_x"), py::arg("h0"), py::arg("c0"), py::arg("wh"),
+          py::arg("stash") = true);
     m.def("lstm_seq_bwd", &lstm_seq_bwd, "fused LSTM sequence backward (K2)");
     m.def("td_targets", &td_targets, "fused n-step TD targets + priorities (K3)",
           py::arg("rewards"), py::arg("dones"), py::arg("q_boot"), py::arg("q_online"),

@@ -86,7 +86,7 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                 DT* __restrict__ c_final,        // [B, H]
                 DT* __restrict__ gates_act,      // [T, B, 4H] post-act i,f,g,o
                 float* __restrict__ c_seq,       // [T, B, H]
-                int T, int B, int H) {
+                int T, int B, int H, int stash) {  // stash=0: inference (no bwd)
     using F = Frag<DT>;
     // TWO h tiles (double buffer): read h(t-1) from buf[t&1], write h(t) to
     // buf[(t+1)&1] -> one barrier per timestep instead of two.
@@ -203,11 +203,13 @@ lstm_fwd_kernel(const DT* __restrict__ gates_x,  // [T, B, 4H] (x@Wx^T + b)
                     st_lds<DT>(wr_buf, swz(row * rowbytes + col * sizeof(DT), row), hn);
                     if (grow < B) {
                         st_f(&hs_t[(long)grow * H + col], hn);
-                        cs_t[(long)grow * H + col] = c;
-                        st_f(&ga_t[(long)grow * H4 + 0 * H + col], gi);
-                        st_f(&ga_t[(long)grow * H4 + 1 * H + col], gf);
-                        st_f(&ga_t[(long)grow * H4 + 2 * H + col], gg);
-                        st_f(&ga_t[(long)grow * H4 + 3 * H + col], go);
+                        if (stash) {  // wave-uniform: skipped for burn-in/target unrolls
+                            cs_t[(long)grow * H + col] = c;
+                            st_f(&ga_t[(long)grow * H4 + 0 * H + col], gi);
+                            st_f(&ga_t[(long)grow * H4 + 1 * H + col], gf);
+                            st_f(&ga_t[(long)grow * H4 + 2 * H + col], gg);
+                            st_f(&ga_t[(long)grow * H4 + 3 * H + col], go);
+                        }
                     }
                 }
             }
@@ -409,7 +411,7 @@ static inline int pick_geom(int H, int* threads, int* hf) {
 extern "C" int launch_lstm_fwd_bf16(
     const void* gates_x, const void* h0, const void* c0, const void* wh,
     void* h_seq, void* c_final, void* gates_act, float* c_seq,
-    int T, int B, int H, hipStream_t stream) {
+    int T, int B, int H, int stash, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
     size_t lds = (size_t)2 * 16 * H * sizeof(__hip_bfloat16);
@@ -418,14 +420,14 @@ extern "C" int launch_lstm_fwd_bf16(
             (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
             (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh,
             (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
-            (__hip_bfloat16*)gates_act, c_seq, T, B, H)));
+            (__hip_bfloat16*)gates_act, c_seq, T, B, H, stash)));
     return 0;
 }
 
 extern "C" int launch_lstm_fwd_f32(
     const void* gates_x, const void* h0, const void* c0, const void* wh,
     void* h_seq, void* c_final, void* gates_act, float* c_seq,
-    int T, int B, int H, hipStream_t stream) {
+    int T, int B, int H, int stash, hipStream_t stream) {
     int blocks = (B + 15) / 16, threads, hf;
     if (pick_geom(H, &threads, &hf)) return -1;
     size_t lds = (size_t)2 * 16 * H * sizeof(float);
@@ -433,7 +435,7 @@ extern "C" int launch_lstm_fwd_f32(
         (lstm_fwd_kernel<float, HF><<<dim3(blocks), dim3(threads), lds, stream>>>(
             (const float*)gates_x, (const float*)h0, (const float*)c0,
             (const float*)wh, (float*)h_seq, (float*)c_final,
-            (float*)gates_act, c_seq, T, B, H)));
+            (float*)gates_act, c_seq, T, B, H, stash)));
     return 0;
 }
 

@@ -81,6 +81,17 @@ def lstm_seq(
     Returns (h_seq [T,B,H], hT [B,H], cT [B,H]).
     """
     if x.is_cuda:
+        if not torch.is_grad_enabled():
+            # burn-in / target unrolls: no-stash fast path (no activation
+            # stores, nothing saved for backward)
+            T, B, I = x.shape
+            H = h0.shape[1]
+            ext = hip_ext()
+            gates_x = torch.addmm(b, x.reshape(T * B, I), wx.t()).view(T, B, 4 * H)
+            h_seq, cT, _, _ = ext.lstm_seq_fwd(
+                gates_x.contiguous(), h0.contiguous(), c0.contiguous(),
+                wh.contiguous(), False)
+            return h_seq, h_seq[-1], cT
         h_seq, cT = _FusedLSTMSeq.apply(x, h0, c0, wx, wh, b)
         return h_seq, h_seq[-1], cT
     return lstm_seq_ref(x, h0, c0, wx, wh, b)

@@ -280,3 +280,33 @@ def test_learner_step_gpu_bf16():
     stats = lrn.train_step(batch)
     assert np.isfinite(stats.critic_loss) and np.isfinite(stats.actor_loss)
     assert any(not torch.equal(a, b) for a, b in zip(p0, lrn.actor.parameters()))
+
+
+def test_sumtree_stress_consistency():
+    """Race-validation stress: hammer interleaved add/sample/update
+    from the real call topology; root must equal the leaf sum throughout."""
+    cap = 4096
+    tree = torch.zeros(2 * cap, dtype=torch.float32, device="cuda")
+    g = torch.Generator().manual_seed(3)
+    e = ext()
+    for round_ in range(200):
+        n_new = int(torch.randint(1, 64, (1,), generator=g))
+        idx = torch.randperm(cap, generator=g)[:n_new].cuda()
+        pri = (torch.rand(n_new, generator=g) + 0.01).cuda()
+        e.sumtree_update(tree, cap, idx, pri)
+        if round_ % 3 == 0:
+            sidx, _ = e.sumtree_sample(tree, cap, 256, 7, round_)
+            upd = (torch.rand(256, generator=g) + 0.01).cuda()
+            # duplicate slots in one sampled batch: deduplicate like the
+            # learner's priority write-back contract requires
+            uniq, inv = torch.unique(sidx, return_inverse=True)
+            first = torch.full((uniq.numel(),), -1, dtype=torch.long, device="cuda")
+            first.scatter_(0, inv, torch.arange(256, device="cuda"))
+            e.sumtree_update(tree, cap, uniq, upd[first])
+        if round_ % 25 == 0:
+            t = tree.cpu()
+            assert torch.isclose(t[1], t[cap:].sum(), rtol=1e-4), round_
+            # every internal node equals the sum of its children
+            nodes = t[1:cap]
+            kids = t[2::2][:cap - 1] + t[3::2][:cap - 1]
+            assert torch.allclose(nodes, kids, rtol=1e-4, atol=1e-5), round_

@@ -20,7 +20,7 @@ CSRC = os.path.join(PKG, "csrc")
 OUT = os.path.join(PKG, "_r2d2dpg_hip.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-HIP_SOURCES = ["lstm.hip", "td_polyak.hip", "sumtree.hip", "replay_pack.hip"]
+HIP_SOURCES = ["lstm.hip", "lstm_px.hip", "td_polyak.hip", "sumtree.hip", "replay_pack.hip"]
 CPP_SOURCES = ["bindings.cpp"]
 
 

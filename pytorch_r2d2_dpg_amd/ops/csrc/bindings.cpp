@@ -8,6 +8,8 @@
 #include <hip/hip_runtime.h>
 
 #include <tuple>
+#include <unordered_map>
+#include <ATen/hip/HIPContext.h>
 
 // ---- extern "C" launchers from the .hip TUs ------------------------------
 extern "C" int launch_lstm_fwd_bf16(const void*, const void*, const void*, const void*,
@@ -34,11 +36,36 @@ extern "C" void launch_sumtree_update(float*, int, const long*, const float*, in
 extern "C" void launch_fp8_encode(const float*, unsigned char*, long, float, hipStream_t);
 extern "C" void launch_fp8_gather_decode(const unsigned char*, const long*, void*, int,
                                          long, int, float, hipStream_t);
+extern "C" int launch_lstm_fwd_px_bf16(const void*, const void*, const void*, const void*,
+                                       void*, void*, void*, float*, unsigned long long*,
+                                       unsigned int*, int, int, int, int, hipStream_t);
 
 namespace {
 
 hipStream_t cur_stream() {
     return at::hip::getCurrentHIPStream().stream();
+}
+
+// Per-device error word for the column-split kernel's bounded spins.
+torch::Tensor& px_err_word(const torch::Device& dev) {
+    static std::unordered_map<int, torch::Tensor> words;
+    int idx = dev.index() < 0 ? 0 : dev.index();
+    auto it = words.find(idx);
+    if (it == words.end())
+        it = words.emplace(idx, torch::zeros({1}, torch::dtype(torch::kInt32).device(dev))).first;
+    return it->second;
+}
+
+// Read-and-clear the column-split spin-timeout error word (0 = healthy).
+int64_t lstm_px_error() {
+    int64_t total = 0;
+    // check word on the current device only (tests run single-GPU)
+    if (torch::cuda::is_available()) {
+        auto& w = px_err_word(torch::Device(torch::kCUDA, at::cuda::current_device()));
+        total = w.item<int32_t>();
+        w.zero_();
+    }
+    return total;
 }
 
 // Pack a [R, C] weight into the MFMA-fragment-major layout the LSTM kernels
@@ -86,12 +113,27 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
                               gates_x.options().dtype(torch::kFloat32));
 
     int rc;
-    if (dt == torch::kBFloat16)
-        rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
-                                  wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
-                                  gates_act.data_ptr(),
-                                  stash ? c_seq.data_ptr<float>() : nullptr,
-                                  (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
+    if (dt == torch::kBFloat16) {
+        // column-split kernel first (CG-way less Wh streaming per CU); it
+        // declines geometries where its blocks can't all be co-resident
+        const int RT = (int)((B + 15) / 16);
+        auto hx = torch::zeros({2LL, RT, 16, H / 2},
+                               gates_x.options().dtype(torch::kInt64));
+        auto& err = px_err_word(gates_x.device());
+        rc = launch_lstm_fwd_px_bf16(
+            gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(), wh_p.data_ptr(),
+            h_seq.data_ptr(), c_final.data_ptr(), gates_act.data_ptr(),
+            stash ? c_seq.data_ptr<float>() : nullptr,
+            (unsigned long long*)hx.data_ptr<int64_t>(),
+            (unsigned int*)err.data_ptr<int32_t>(),
+            (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
+        if (rc != 0)
+            rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
+                                      wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
+                                      gates_act.data_ptr(),
+                                      stash ? c_seq.data_ptr<float>() : nullptr,
+                                      (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
+    }
     else
         rc = launch_lstm_fwd_f32(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
                                  wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
@@ -255,4 +297,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fp8_gather_decode", &fp8_gather_decode, "fp8 fused gather+decode (K7)",
           py::arg("store"), py::arg("idx"), py::arg("out_dtype"),
           py::arg("inv_scale") = 1.0);
+    m.def("lstm_px_error", &lstm_px_error,
+          "read-and-clear the column-split LSTM spin-timeout error word");
 }

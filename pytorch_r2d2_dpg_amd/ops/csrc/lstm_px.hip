@@ -34,9 +34,10 @@ DEV_INLINE unsigned px_swz(unsigned byte_off, int row) {
     return byte_off ^ ((unsigned)(row & 7) << 4);
 }
 
-// HFRAG fixed at 2 (two 16-col fragments per gate per wave).
-template <int DUMMY>
-__global__ void __launch_bounds__(1024)
+// One 16-col fragment per gate per wave; CG is compile-time so the sweep's
+// per-thread granule count (2*CG) is a static register array.
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
 lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                    const __hip_bfloat16* __restrict__ h0,       // [B,H]
                    const __hip_bfloat16* __restrict__ c0,       // [B,H]
@@ -47,7 +48,7 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                    float* __restrict__ c_seq,                   // [T,B,H]
                    unsigned long long* __restrict__ hx,         // [2][RT][16][H/2] zeroed
                    unsigned int* __restrict__ err,
-                   int T, int B, int H, int CG, int stash) {
+                   int T, int B, int H, int stash) {
     extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x h tile [16][H] bf16
 
     const int tid = threadIdx.x;
@@ -88,32 +89,49 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     const long slot_sz = (long)gridDim.x * 16 * (H / 2);
 
     for (int t = 0; t < T; ++t) {
-        // ---- sweep peers' h(t-1) granules into LDS buf[t&1] (skip t=0: h0
-        // already local). Own cols were written to LDS at the epilogue.
+        // ---- sweep ALL h(t-1) granules (peers' and own) into LDS buf[t&1]
+        // (skip t=0: h0 already local). Batched: each thread issues its 2*CG
+        // granule loads unconditionally (one L2 round-trip covers all),
+        // then re-polls only the not-yet-tagged ones.
         if (t > 0) {
             char* bld = smem + (t & 1) * tile_bytes;
-            const unsigned tag = (unsigned)t;  // epoch t = h(t-1), tags start at 1
+            const unsigned tag = (unsigned)t;  // epoch t = h(t-1), tags from 1
             gu64* slot = hx_g + ((t - 1) & 1) * slot_sz + rt_off;
-            // granules for columns OUTSIDE [cg*HQg, cg*HQg+HQg)
-            const int own_lo = cg * HQg / 2, own_hi = (cg + 1) * HQg / 2;
-            const int row_gr = H / 2;
-            const int total = 16 * row_gr;
-            for (int e = tid; e < total; e += (int)blockDim.x) {
-                int r = e / row_gr, gcol = e - r * row_gr;
-                if (gcol >= own_lo && gcol < own_hi) continue;
-                unsigned long long v = __hip_atomic_load(&slot[e], RLX_AGENT);
-                unsigned spins = 0;
-                while ((unsigned)(v >> 32) != tag) {
-                    if (++spins > 100000000u) {  // bounded spin: bail, flag error
-                        __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
-                        goto bail;
-                    }
-                    __builtin_amdgcn_s_sleep(2);
-                    v = __hip_atomic_load(&slot[e], RLX_AGENT);
+            constexpr int K = 2 * CG;          // 8H granules / (4H/CG threads)
+            const int bd = (int)blockDim.x;
+            const int sh = 31 - __builtin_clz((unsigned)(H >> 1));  // log2(H/2)
+            unsigned long long v[K];
+#pragma unroll
+            for (int i = 0; i < K; ++i)
+                v[i] = __hip_atomic_load(&slot[tid + i * bd], RLX_AGENT);
+            unsigned pending = 0;
+#pragma unroll
+            for (int i = 0; i < K; ++i)
+                pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
+            unsigned spins = 0;
+            while (pending) {
+                if (++spins > 40000000u) {  // bounded spin: bail, flag error
+                    __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
+                    goto bail;
                 }
-                unsigned lo = (unsigned)v;
+                __builtin_amdgcn_s_sleep(1);
+#pragma unroll
+                for (int i = 0; i < K; ++i)
+                    if (pending & (1u << i)) {
+                        unsigned long long nv =
+                            __hip_atomic_load(&slot[tid + i * bd], RLX_AGENT);
+                        if ((unsigned)(nv >> 32) == tag) {
+                            v[i] = nv;
+                            pending &= ~(1u << i);
+                        }
+                    }
+            }
+#pragma unroll
+            for (int i = 0; i < K; ++i) {
+                int e = tid + i * bd;
+                int r = e >> sh, gcol = e & ((H >> 1) - 1);
                 *reinterpret_cast<unsigned*>(
-                    &bld[px_swz(r * rowbytes + gcol * 4, r)]) = lo;  // 2 bf16
+                    &bld[px_swz(r * rowbytes + gcol * 4, r)]) = (unsigned)v[i];
             }
         }
         __syncthreads();
@@ -123,7 +141,6 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
         __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
         float* cs_t = c_seq + (long)t * B * H;
         char* rd_buf = smem + (t & 1) * tile_bytes;
-        char* wr_buf = smem + ((t + 1) & 1) * tile_bytes;
 
         f32x4 acc[4];
 #pragma unroll
@@ -168,8 +185,6 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 float hn = go * px_tanh(c);
                 c_reg[r] = c;
                 unsigned short hb = f32_to_bf16_bits(hn);
-                *reinterpret_cast<__hip_bfloat16*>(
-                    &wr_buf[px_swz(row * rowbytes + col * 2, row)]) = f2bf(hn);
                 // pair columns (even, odd) into one tagged granule
                 unsigned other = __shfl_xor((unsigned)hb, 1);
                 if ((lane & 1) == 0) {
@@ -208,6 +223,7 @@ extern "C" int launch_lstm_fwd_px_bf16(
     int T, int B, int H, int stash, hipStream_t stream) {
     int RT = (B + 15) / 16;
     // largest CG with co-residency margin and >= 2 fragments per gate
+    if (H & (H - 1)) return -1;              // sweep indexing needs pow2 H
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;          // wave owns 16 cols
@@ -221,10 +237,30 @@ extern "C" int launch_lstm_fwd_px_bf16(
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG);
     size_t lds = (size_t)2 * 16 * H * 2;
-    lstm_fwd_px_kernel<0><<<grid, dim3(waves * 64), lds, stream>>>(
-        (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
-        (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
-        (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
-        (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, CG, stash);
+    switch (CG) {
+        case 2:
+            lstm_fwd_px_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(
+                (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
+                (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
+                (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+            break;
+        case 4:
+            lstm_fwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(
+                (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
+                (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
+                (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+            break;
+        case 8:
+            lstm_fwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(
+                (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
+                (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
+                (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+            break;
+        default:
+            return -1;
+    }
     return 0;
 }

@@ -11,12 +11,13 @@ import torch
 from pytorch_r2d2_dpg_amd.config import Config
 
 
-@pytest.mark.timeout(600)
+@pytest.mark.timeout(900)
 def test_pendulum_learns_single_process():
-    """Config-1 smoke: loss finite, value estimates move, policy improves vs
-    the random-init policy on fixed eval seeds (cheap proxy for 'learning
-    works' that stays CI-sized; the full -200 return criterion is the
-    longer-horizon config in configs/pendulum_cpu.toml)."""
+    """Config-1 learning smoke (BASELINE.json config 1): 1500 learner steps of
+    the shipped pendulum config must clearly beat the random policy on fixed
+    eval seeds. Calibrated: this setup reaches eval ~-790 at 1500 steps, -360
+    at 2500, ~-180 (solved) at 4000; a random policy scores ~-1300..-1500.
+    Threshold -1000 leaves seed margin at CI length."""
     from pytorch_r2d2_dpg_amd.actor import ActorWorker
     from pytorch_r2d2_dpg_amd.trainer import train_single_process
 
@@ -24,30 +25,18 @@ def test_pendulum_learns_single_process():
     cfg = Config.from_toml(os.path.join(os.path.dirname(__file__), "..",
                                         "configs", "pendulum_cpu.toml"))
     cfg.learner.checkpoint_every = 0
-    cfg.learner.sync_every = 25
+    cfg.learner.sync_every = 20
 
-    def eval_policy(actor_sd, critic_sd, episodes=5):
-        w = ActorWorker(cfg, actor_id=0, eval_mode=True)
-        w.load_weights(actor_sd, critic_sd)
-        w.env.seed(12345)
-        for _ in range(episodes):
-            w.rollout(cfg.env.time_limit)
-        return float(np.mean(w.episode_returns[:episodes]))
-
-    lrn, actor_w = train_single_process(cfg, max_steps=250, env_steps_per_update=60)
+    lrn, actor_w = train_single_process(cfg, max_steps=1500, env_steps_per_update=40)
     sd = lrn.actor_state_for_sync()
-    ret_trained = eval_policy(sd["actor"], sd["critic"])
-
-    torch.manual_seed(1)
-    fresh = ActorWorker(cfg, actor_id=0, eval_mode=True)
-    fresh.env.seed(12345)
+    ev = ActorWorker(cfg, actor_id=0, eval_mode=True)
+    ev.load_weights(sd["actor"], sd["critic"])
+    ev.env.seed(12345)
     for _ in range(5):
-        fresh.rollout(cfg.env.time_limit)
-    ret_fresh = float(np.mean(fresh.episode_returns[:5]))
-
-    # trained should beat (or at least not be much worse than) random-init
-    assert np.isfinite(ret_trained)
-    assert ret_trained > ret_fresh - 100.0, (ret_trained, ret_fresh)
+        ev.rollout(cfg.env.time_limit)
+    ret = float(np.mean(ev.episode_returns[:5]))
+    assert np.isfinite(ret)
+    assert ret > -1000.0, f"no learning: eval return {ret}"
 
 
 @pytest.mark.timeout(300)

@@ -39,6 +39,10 @@ extern "C" void launch_fp8_gather_decode(const unsigned char*, const long*, void
 extern "C" int launch_lstm_fwd_px_bf16(const void*, const void*, const void*, const void*,
                                        void*, void*, void*, float*, unsigned long long*,
                                        unsigned int*, int, int, int, int, hipStream_t);
+extern "C" int launch_lstm_bwd_px_bf16(const void*, const void*, const void*, const void*,
+                                       const float*, const void*, const void*, void*, void*,
+                                       void*, unsigned char*, unsigned long long*,
+                                       unsigned int*, int, int, int, hipStream_t);
 
 namespace {
 
@@ -168,12 +172,26 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
     auto dc0 = torch::empty({B, H}, dh_seq.options());
 
     int rc;
-    if (dt == torch::kBFloat16)
-        rc = launch_lstm_bwd_bf16(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
-                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),
-                                  c0.data_ptr(), whT_p.data_ptr(), dgates.data_ptr(),
-                                  dh0.data_ptr(), dc0.data_ptr(),
-                                  (int)T, (int)B, (int)H, cur_stream());
+    if (dt == torch::kBFloat16) {
+        const int RT = (int)((B + 15) / 16);
+        auto slab = torch::empty({2LL, RT, 16 * 4 * H * 2},
+                                 dh_seq.options().dtype(torch::kUInt8));
+        auto flags = torch::zeros({2LL, RT, 8}, dh_seq.options().dtype(torch::kInt64));
+        auto& err = px_err_word(dh_seq.device());
+        rc = launch_lstm_bwd_px_bf16(
+            dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(), gates_act.data_ptr(),
+            c_seq.data_ptr<float>(), c0.data_ptr(), whT_p.data_ptr(),
+            dgates.data_ptr(), dh0.data_ptr(), dc0.data_ptr(),
+            slab.data_ptr<uint8_t>(), (unsigned long long*)flags.data_ptr<int64_t>(),
+            (unsigned int*)err.data_ptr<int32_t>(),
+            (int)T, (int)B, (int)H, cur_stream());
+        if (rc != 0)
+            rc = launch_lstm_bwd_bf16(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
+                                      gates_act.data_ptr(), c_seq.data_ptr<float>(),
+                                      c0.data_ptr(), whT_p.data_ptr(), dgates.data_ptr(),
+                                      dh0.data_ptr(), dc0.data_ptr(),
+                                      (int)T, (int)B, (int)H, cur_stream());
+    }
     else
         rc = launch_lstm_bwd_f32(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
                                  gates_act.data_ptr(), c_seq.data_ptr<float>(),

@@ -110,9 +110,13 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 40000000u) {  // bounded spin: bail, flag error
+                if (++spins > 200000000u) {
+                    // bounded spin: flag the error and PROCEED with stale
+                    // data — control flow stays convergent so no thread
+                    // ever skips a barrier (a diverged exit deadlocks the
+                    // block and hangs the device)
                     __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
-                    goto bail;
+                    break;
                 }
                 __builtin_amdgcn_s_sleep(1);
 #pragma unroll
@@ -205,9 +209,8 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 }
             }
         }
-        __syncthreads();  // LDS wr_buf complete before next sweep reads buf
+        __syncthreads();  // next sweep writes the other LDS buffer
     }
-bail:
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         int grow = brow0 + c_row0 + r;
@@ -262,5 +265,241 @@ extern "C" int launch_lstm_fwd_px_bf16(
         default:
             return -1;
     }
+    return 0;
+}
+
+// ---------------------------------------------------------------- backward
+// Column-split backward. Block (rt, cg) owns h-cols [cg*HQg, +HQg): computes
+// dgates for its 4*HQg gate columns lane-locally, publishes them as an sc1
+// SLAB (the swizzled LDS-tile byte image) + one tagged flag per block per
+// step (flag stored only after every wave drained its sc1 stores — G16 R1),
+// sweeps peers' slabs into the LDS tile, then runs the dh recurrence MFMA
+// with K = 4H against its whT row slice (2MB/CG streamed per step).
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
+                   const __hip_bfloat16* __restrict__ dhT,      // [B,H]
+                   const __hip_bfloat16* __restrict__ dcT,      // [B,H]
+                   const __hip_bfloat16* __restrict__ gates_act,// [T,B,4H]
+                   const float* __restrict__ c_seq,             // [T,B,H]
+                   const __hip_bfloat16* __restrict__ c0,       // [B,H]
+                   const __hip_bfloat16* __restrict__ whT,      // packed [H/16][4H/32][64][8]
+                   __hip_bfloat16* __restrict__ dgates,         // [T,B,4H]
+                   __hip_bfloat16* __restrict__ dh0,            // [B,H]
+                   __hip_bfloat16* __restrict__ dc0,            // [B,H]
+                   unsigned char* __restrict__ slab,            // [2][RT][16*4H*2] bytes
+                   unsigned long long* __restrict__ flags,      // [2][RT][CG] zeroed
+                   unsigned int* __restrict__ err,
+                   int T, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x dg tile [16][4H]
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int rt = blockIdx.x;
+    const int cg = blockIdx.y;
+    const int brow0 = rt * 16;
+    const int HQg = H / CG;
+    const int col_base = cg * HQg + wid * 16;
+    const int H4 = 4 * H;
+    const int rowbytes4 = H4 * 2;
+    const unsigned tile_bytes = 16u * (unsigned)rowbytes4;
+
+    const int a_row = lane & 15;
+    const int a_kgrp = (lane >> 4) * 8;
+    const int c_col = lane & 15;
+    const int c_row0 = (lane >> 4) * 4;
+
+    float dh_reg[4], dc_reg[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        bool ok = grow < B;
+        dh_reg[r] = ok ? bf2f(dhT[(long)grow * H + col]) : 0.0f;
+        dc_reg[r] = ok ? bf2f(dcT[(long)grow * H + col]) : 0.0f;
+    }
+
+    typedef __attribute__((address_space(1))) unsigned long long gu64b;
+    const long slab_tile = (long)tile_bytes;
+    const long slab_slot = (long)gridDim.x * slab_tile;
+    gu64b* flg = (gu64b*)flags;
+    const int own_byte_lo = (cg * HQg) * 2;   // within one gate's row segment
+
+    for (int t = T - 1; t >= 0; --t) {
+        const unsigned tag = (unsigned)(T - t);
+        char* buf = smem + (t & 1) * tile_bytes;
+        const __hip_bfloat16* dh_t = dh_seq + (long)t * B * H;
+        const __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
+        const float* cs_t = c_seq + (long)t * B * H;
+        const float* cs_p = (t > 0) ? c_seq + (long)(t - 1) * B * H : nullptr;
+        __hip_bfloat16* dg_t = dgates + (long)t * B * H4;
+
+        // ---- lane-local dgates for own columns -> LDS own region + global
+        {
+            const int col = col_base + c_col;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = c_row0 + r;
+                int grow = brow0 + row;
+                float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
+                if (grow < B) {
+                    float dht = dh_reg[r] + bf2f(dh_t[(long)grow * H + col]);
+                    float gi = bf2f(ga_t[(long)grow * H4 + 0 * H + col]);
+                    float gf = bf2f(ga_t[(long)grow * H4 + 1 * H + col]);
+                    float gg = bf2f(ga_t[(long)grow * H4 + 2 * H + col]);
+                    float go = bf2f(ga_t[(long)grow * H4 + 3 * H + col]);
+                    float ct = cs_t[(long)grow * H + col];
+                    float cp = (t > 0) ? cs_p[(long)grow * H + col]
+                                       : bf2f(c0[(long)grow * H + col]);
+                    float tc = px_tanh(ct);
+                    float do_ = dht * tc;
+                    float dc = dht * go * (1.0f - tc * tc) + dc_reg[r];
+                    dgi = dc * gg * gi * (1.0f - gi);
+                    dgf = dc * cp * gf * (1.0f - gf);
+                    dgg = dc * gi * (1.0f - gg * gg);
+                    dgo = do_ * go * (1.0f - go);
+                    dc_reg[r] = dc * gf;
+                    dg_t[(long)grow * H4 + 0 * H + col] = f2bf(dgi);
+                    dg_t[(long)grow * H4 + 1 * H + col] = f2bf(dgf);
+                    dg_t[(long)grow * H4 + 2 * H + col] = f2bf(dgg);
+                    dg_t[(long)grow * H4 + 3 * H + col] = f2bf(dgo);
+                } else {
+                    dc_reg[r] = 0.0f;
+                }
+                const float vals[4] = {dgi, dgf, dgg, dgo};
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+                    *reinterpret_cast<__hip_bfloat16*>(
+                        &buf[px_swz(row * rowbytes4 + (g * H + col) * 2, row)]) =
+                        f2bf(vals[g]);
+            }
+        }
+        __syncthreads();
+
+        // ---- publish own slab region (sc1 16B), drain, flag
+        {
+            unsigned char* slab_rt = slab + (t & 1) * slab_slot + (long)rt * slab_tile;
+            auto rsrc = __builtin_amdgcn_make_buffer_rsrc(slab_rt, 0, (int)tile_bytes,
+                                                          0x00020000);
+            // own bytes: 16 rows x 4 gates x HQg*2 bytes, each range 128B-aligned
+            // and XOR-swizzle-closed, so raw byte copy preserves the image
+            const int own_bytes = HQg * 2;
+            const int chunks = 16 * 4 * (own_bytes / 16);
+            for (int ch = tid; ch < chunks; ch += (int)blockDim.x) {
+                int per_row = 4 * (own_bytes / 16);
+                int row = ch / per_row;
+                int rem = ch - row * per_row;
+                int g = rem / (own_bytes / 16);
+                int o16 = rem - g * (own_bytes / 16);
+                unsigned off = (unsigned)(row * rowbytes4 + g * H * 2 + own_byte_lo + o16 * 16);
+                f32x4 v = *reinterpret_cast<const f32x4*>(&buf[off]);
+                __builtin_amdgcn_raw_buffer_store_b128(v, rsrc, (int)off, 0, 16);
+            }
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __syncthreads();
+            if (tid == 0)
+                __hip_atomic_store(&flg[((t & 1) * gridDim.x + rt) * CG + cg],
+                                   ((unsigned long long)tag << 32) | 1u, RLX_AGENT);
+        }
+
+        // ---- wait peers' flags, then sweep their slab regions into LDS
+        {
+            if (tid < CG && tid != cg) {
+                gu64b* f = &flg[((t & 1) * gridDim.x + rt) * CG + tid];
+                unsigned long long v = __hip_atomic_load(f, RLX_AGENT);
+                unsigned spins = 0;
+                while ((unsigned)(v >> 32) != tag) {
+                    if (++spins > 200000000u) {  // convergent bail (see fwd)
+                        __hip_atomic_store((gu32*)err, 0x10000u + (unsigned)t, RLX_AGENT);
+                        break;
+                    }
+                    __builtin_amdgcn_s_sleep(1);
+                    v = __hip_atomic_load(f, RLX_AGENT);
+                }
+            }
+            __syncthreads();
+            unsigned char* slab_rt = slab + (t & 1) * slab_slot + (long)rt * slab_tile;
+            auto rsrc = __builtin_amdgcn_make_buffer_rsrc(slab_rt, 0, (int)tile_bytes,
+                                                          0x00020000);
+            const int own_bytes = HQg * 2;
+            const int per_gate16 = (H * 2 - own_bytes) / 16;  // peer bytes per gate-row
+            const int chunks = 16 * 4 * per_gate16;
+            for (int ch = tid; ch < chunks; ch += (int)blockDim.x) {
+                int per_row = 4 * per_gate16;
+                int row = ch / per_row;
+                int rem = ch - row * per_row;
+                int g = rem / per_gate16;
+                int o16 = rem - g * per_gate16;
+                int byte_in_gate = o16 * 16 + (o16 * 16 >= own_byte_lo ? own_bytes : 0);
+                unsigned off = (unsigned)(row * rowbytes4 + g * H * 2 + byte_in_gate);
+                f32x4 v = __builtin_amdgcn_raw_buffer_load_b128(rsrc, (int)off, 0, 16);
+                *reinterpret_cast<f32x4*>(&buf[off]) = v;
+            }
+        }
+        __syncthreads();
+
+        // ---- dh recurrence: dh_prev(own cols) = dgates(all 4H) @ Wh
+        f32x4 acc = {0, 0, 0, 0};
+        for (int kk = 0; kk < H4; kk += 32) {
+            int k = kk + a_kgrp;
+            short8 a_frag = *reinterpret_cast<const short8*>(
+                &buf[px_swz(a_row * rowbytes4 + k * 2, a_row)]);
+            int nb = (col_base >> 4);
+            long off = (((long)nb * (H4 >> 5) + (kk >> 5)) * 64 + lane) * 8;
+            short8 b_frag = *reinterpret_cast<const short8*>(&whT[off]);
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) dh_reg[r] = acc[r];
+        __syncthreads();  // LDS buf reused two iterations later (double buffer)
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        if (grow < B) {
+            dh0[(long)grow * H + col] = f2bf(dh_reg[r]);
+            dc0[(long)grow * H + col] = f2bf(dc_reg[r]);
+        }
+    }
+}
+
+extern "C" int launch_lstm_bwd_px_bf16(
+    const void* dh_seq, const void* dhT, const void* dcT, const void* gates_act,
+    const float* c_seq, const void* c0, const void* whT_packed,
+    void* dgates, void* dh0, void* dc0,
+    unsigned char* slab, unsigned long long* flags, unsigned int* err,
+    int T, int B, int H, hipStream_t stream) {
+    int RT = (B + 15) / 16;
+    if (H & (H - 1)) return -1;
+    if ((size_t)2 * 16 * 4 * H * 2 > 160 * 1024) return -1;  // LDS double buffer
+    int CG = 0;
+    for (int c : {8, 4, 2}) {
+        if (H % (16 * c)) continue;
+        int w = (H / c) / 16;
+        if (w < 2 || w > 16) continue;
+        if (RT * c > 224) continue;
+        CG = c;
+        break;
+    }
+    if (CG == 0) return -1;
+    int waves = (H / CG) / 16;
+    dim3 grid(RT, CG);
+    size_t lds = (size_t)2 * 16 * 4 * H * 2;
+#define BWD_PX_CALL(CGV)                                                        \
+    lstm_bwd_px_kernel<CGV><<<grid, dim3(waves * 64), lds, stream>>>(           \
+        (const __hip_bfloat16*)dh_seq, (const __hip_bfloat16*)dhT,              \
+        (const __hip_bfloat16*)dcT, (const __hip_bfloat16*)gates_act, c_seq,    \
+        (const __hip_bfloat16*)c0, (const __hip_bfloat16*)whT_packed,           \
+        (__hip_bfloat16*)dgates, (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0,    \
+        slab, flags, err, T, B, H)
+    switch (CG) {
+        case 2: BWD_PX_CALL(2); break;
+        case 4: BWD_PX_CALL(4); break;
+        case 8: BWD_PX_CALL(8); break;
+        default: return -1;
+    }
+#undef BWD_PX_CALL
     return 0;
 }

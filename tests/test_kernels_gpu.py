@@ -312,7 +312,7 @@ def test_sumtree_stress_consistency():
             assert torch.allclose(nodes, kids, rtol=1e-4, atol=1e-5), round_
 
 
-@pytest.mark.parametrize("shape", [(256, 512, 40), (33, 256, 12), (448, 512, 7)])
+@pytest.mark.parametrize("shape", [(256, 512, 40), (33, 256, 12), (448, 512, 7), (512, 512, 6)])
 def test_lstm_fwd_px_vs_ref(shape):
     """Column-split forward (K1-px): numerics vs fp32 reference + no spin
     timeouts (the bounded-spin error word must stay 0)."""
@@ -331,3 +331,24 @@ def test_lstm_fwd_px_vs_ref(shape):
     assert cerr < 0.25, f"px bf16 c_final err {cerr}"
     # stash consistency: gates/c_seq shapes filled
     assert ga.shape[0] == T and cs.shape[0] == T
+
+
+@pytest.mark.parametrize("shape", [(256, 512, 13), (48, 256, 9), (512, 512, 5)])
+def test_lstm_bwd_px_vs_ref(shape):
+    """Column-split backward (K2-px): autograd grads vs fp32 reference."""
+    B, H, T = shape
+    I = 24
+    cpu, gpu = _mk_lstm(T, B, I, H, torch.bfloat16, seed=21)
+    cpu = tuple(t.clone().requires_grad_(True) for t in cpu)
+    gpu = tuple(t.clone().requires_grad_(True) for t in gpu)
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq
+    h_ref, _, cT_ref = lstm_seq_ref(*cpu)
+    (h_ref.square().mean() + cT_ref.abs().mean()).backward()
+    h_gpu, _, cT_gpu = lstm_seq(*gpu)
+    (h_gpu.square().mean() + cT_gpu.abs().mean()).backward()
+    torch.cuda.synchronize()
+    assert ext().lstm_px_error() == 0, "px bwd hit its spin bound"
+    for name, rc_, gc in zip(("x", "h0", "c0", "wx", "wh", "b"), cpu, gpu):
+        num = (rc_.grad - gc.grad.float().cpu()).norm()
+        den = rc_.grad.norm().clamp_min(1e-3)
+        assert num / den < 0.09, f"d{name} rel-norm err {num / den}"

@@ -479,6 +479,10 @@ extern "C" int launch_lstm_bwd_px_bf16(
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
+        if (H / c < 64) continue;  // slab regions must be >= 128B: the XOR
+                                   // swizzle permutes 16B blocks within a
+                                   // 128B window, so raw byte-range copies
+                                   // are closed only for >=128B regions
         if (RT * c > 224) continue;
         CG = c;
         break;

@@ -111,6 +111,7 @@ def main() -> int:
         obs_shape=spec.obs_shape, action_dim=spec.action_dim,
         hidden_size=cfg.model.hidden_size, per_alpha=cfg.replay.per_alpha,
         device=device, seed=cfg.learner.seed + 17 * rank,
+        store_dtype=torch.bfloat16 if args.dtype == "bf16" else torch.float32,
     )
     fill_replay_synthetic(replay, spec, cfg, min(2048, cfg.replay.capacity),
                           seed=cfg.learner.seed + 31 * rank)

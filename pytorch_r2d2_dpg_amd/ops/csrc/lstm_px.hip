@@ -291,7 +291,7 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
                    unsigned long long* __restrict__ flags,      // [2][RT][CG] zeroed
                    unsigned int* __restrict__ err,
                    int T, int B, int H) {
-    extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x dg tile [16][4H]
+    extern __shared__ __attribute__((aligned(16))) char smem[];  // dg tile [16][4H]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -328,7 +328,10 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
 
     for (int t = T - 1; t >= 0; --t) {
         const unsigned tag = (unsigned)(T - t);
-        char* buf = smem + (t & 1) * tile_bytes;
+        // SINGLE LDS buffer (64KB at H=512): a double buffer would force one
+        // block/CU and two concurrent bwd kernels would need every CU
+        // co-resident with zero margin — any displaced block then spins out
+        char* buf = smem;
         const __hip_bfloat16* dh_t = dh_seq + (long)t * B * H;
         const __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
         const float* cs_t = c_seq + (long)t * B * H;
@@ -452,7 +455,7 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
         }
 #pragma unroll
         for (int r = 0; r < 4; ++r) dh_reg[r] = acc[r];
-        __syncthreads();  // LDS buf reused two iterations later (double buffer)
+        __syncthreads();  // MFMA reads done before next step's LDS writes
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -473,7 +476,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
     int T, int B, int H, hipStream_t stream) {
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
-    if ((size_t)2 * 16 * 4 * H * 2 > 160 * 1024) return -1;  // LDS double buffer
+    if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;  // keep 2 blocks/CU
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
@@ -490,7 +493,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
     if (CG == 0) return -1;
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG);
-    size_t lds = (size_t)2 * 16 * 4 * H * 2;
+    size_t lds = (size_t)16 * 4 * H * 2;
 #define BWD_PX_CALL(CGV)                                                        \
     lstm_bwd_px_kernel<CGV><<<grid, dim3(waves * 64), lds, stream>>>(           \
         (const __hip_bfloat16*)dh_seq, (const __hip_bfloat16*)dhT,              \

@@ -47,6 +47,7 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
         hidden_size=cfg.model.hidden_size, per_alpha=R.per_alpha, per_beta=R.per_beta,
         device=device, fp8_obs=R.fp8_obs, seed=cfg.learner.seed + 1000 * rank,
         obs_dtype=torch.uint8 if spec.pixel else torch.float32,
+        store_dtype=torch.bfloat16 if cfg.model.dtype == "bf16" else torch.float32,
     )
     learner = Learner(cfg, spec, device=device, dist=dist, replay=replay)
     if resume:

@@ -43,6 +43,7 @@ class GpuSequenceReplay:
         device: str = "cpu",
         obs_dtype: torch.dtype = torch.float32,
         fp8_obs: bool = False,
+        store_dtype: torch.dtype = torch.float32,  # action/state (and float obs)
         seed: int = 0,
     ):
         self.capacity = capacity
@@ -54,14 +55,19 @@ class GpuSequenceReplay:
         self.is_cuda = self.device.type == "cuda"
         self.fp8_obs = fp8_obs and self.is_cuda
         L = stored_len
-        store_obs_dtype = torch.uint8 if self.fp8_obs else obs_dtype
+        if self.fp8_obs:
+            store_obs_dtype = torch.uint8
+        elif obs_dtype == torch.uint8:
+            store_obs_dtype = torch.uint8
+        else:
+            store_obs_dtype = store_dtype
         self.obs = torch.zeros((capacity, L) + tuple(obs_shape), dtype=store_obs_dtype,
                                device=self.device)
-        self.action = torch.zeros((capacity, L, action_dim), dtype=torch.float32,
+        self.action = torch.zeros((capacity, L, action_dim), dtype=store_dtype,
                                   device=self.device)
         self.reward = torch.zeros((capacity, L), dtype=torch.float32, device=self.device)
         self.done = torch.zeros((capacity, L), dtype=torch.float32, device=self.device)
-        self.state = torch.zeros((capacity, 4, hidden_size), dtype=torch.float32,
+        self.state = torch.zeros((capacity, 4, hidden_size), dtype=store_dtype,
                                  device=self.device)
         self.tree = torch.zeros(2 * self.cap2, dtype=torch.float32, device=self.device)
         self._next = 0
@@ -126,10 +132,10 @@ class GpuSequenceReplay:
             self.obs[didx] = enc.view(self.obs[didx].shape)
         else:
             self.obs[didx] = obs.to(self.device, self.obs.dtype, non_blocking=True)
-        self.action[didx] = action.to(self.device, torch.float32, non_blocking=True)
+        self.action[didx] = action.to(self.device, self.action.dtype, non_blocking=True)
         self.reward[didx] = reward.to(self.device, torch.float32, non_blocking=True)
         self.done[didx] = done.to(self.device, torch.float32, non_blocking=True)
-        self.state[didx] = state.to(self.device, torch.float32, non_blocking=True)
+        self.state[didx] = state.to(self.device, self.state.dtype, non_blocking=True)
         if not torch.is_tensor(priority):
             priority = torch.as_tensor(priority)
         p = priority.to(self.device, torch.float32).clamp_min(1e-6) ** self.per_alpha

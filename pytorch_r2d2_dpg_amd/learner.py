@@ -98,7 +98,8 @@ class Learner:
         # stream-overlap machinery (GPU only): the actor chain runs on a side
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
-        if self.device.type == "cuda":
+        import os as _os
+        if self.device.type == "cuda" and not _os.environ.get("R2D2_NO_OVERLAP"):
             self._side_stream = torch.cuda.Stream()
             self._ev_ba = torch.cuda.Event()
             self._ev_bc = torch.cuda.Event()

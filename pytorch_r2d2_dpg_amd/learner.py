@@ -99,7 +99,14 @@ class Learner:
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
         import os as _os
-        if self.device.type == "cuda" and not _os.environ.get("R2D2_NO_OVERLAP"):
+        # Stream overlap pays when the column-split LSTM kernels of the two
+        # chains can co-reside (B <= 256: +20% measured). At larger batches
+        # two spinning px grids contend for CUs and crawl — run sequential
+        # (the px kernels already fill the chip). R2D2_NO_OVERLAP=1 forces off.
+        overlap_ok = (self.device.type == "cuda"
+                      and cfg.learner.batch_size <= 256
+                      and not _os.environ.get("R2D2_NO_OVERLAP"))
+        if overlap_ok:
             self._side_stream = torch.cuda.Stream()
             self._ev_ba = torch.cuda.Event()
             self._ev_bc = torch.cuda.Event()

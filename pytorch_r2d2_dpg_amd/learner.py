@@ -116,6 +116,24 @@ class Learner:
             self._side_stream = None
             self._ev_ba = self._ev_bc = self._ev_cf = self._ev_ab = None
 
+        # optional hipEvent phase timers (SURVEY §5 tracing): R2D2_PHASE_TIMERS=1
+        self._phase_events = None
+        if self.device.type == "cuda" and _os.environ.get("R2D2_PHASE_TIMERS"):
+            self._phase_events = {k: (torch.cuda.Event(enable_timing=True),
+                                      torch.cuda.Event(enable_timing=True))
+                                  for k in ("burn_in", "target", "critic", "actor_opt")}
+
+    def phase_times_ms(self) -> Dict[str, float]:
+        """Elapsed ms per learner phase for the LAST step (synchronizes)."""
+        if not self._phase_events:
+            return {}
+        torch.cuda.synchronize()
+        return {k: s.elapsed_time(e) for k, (s, e) in self._phase_events.items()}
+
+    def _pe(self, key: str, which: int):
+        if self._phase_events:
+            self._phase_events[key][which].record()
+
     # ---------------------------------------------------------- core update
     def _step_core(self, batch: Dict[str, torch.Tensor],
                    update_priorities: bool = True,
@@ -156,6 +174,7 @@ class Learner:
         if overlap:
             s1.wait_stream(main)
 
+        self._pe("burn_in", 0)
         # ---- burn-ins: actor chain on main, critic chain on the side stream
         if bi > 0:
             with torch.no_grad():
@@ -181,6 +200,8 @@ class Learner:
             if overlap:
                 ev_ab.record(s1)
 
+        self._pe("burn_in", 1)
+        self._pe("target", 0)
         # ---- target unroll + critic online fwd (main stream; balances the
         # side stream's actor-chain length)
         with torch.no_grad():
@@ -198,6 +219,8 @@ class Learner:
                 self.gamma, self.n_step, self.eta,
                 rescale=cfg.learner.value_rescale,
             )
+        self._pe("target", 1)
+        self._pe("critic", 0)
         critic_loss = (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
         self.opt_critic.zero_grad(set_to_none=zero_to_none)
         critic_loss.backward()
@@ -209,6 +232,8 @@ class Learner:
             main.wait_event(ev_ab)
         self.opt_critic.step()
 
+        self._pe("critic", 1)
+        self._pe("actor_opt", 0)
         # ---- actor optimizer (grads computed on the side stream)
         for p, g in zip(self._actor_params, actor_grads):
             if p.grad is None:
@@ -226,6 +251,7 @@ class Learner:
             self.polyak_actor(cfg.learner.tau)
             self.polyak_critic(cfg.learner.tau)
 
+        self._pe("actor_opt", 1)
         # ---- priority write-back (stays on-GPU with the K6 tree kernel)
         if update_priorities and self.replay is not None and batch.get("idx") is not None:
             self.replay.update_priorities(batch["idx"], prio)

@@ -99,12 +99,11 @@ class Learner:
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
         import os as _os
-        # Stream overlap pays when the column-split LSTM kernels of the two
-        # chains can co-reside (B <= 256: +20% measured). At larger batches
-        # two spinning px grids contend for CUs and crawl — run sequential
-        # (the px kernels already fill the chip). R2D2_NO_OVERLAP=1 forces off.
+        # Stream overlap hides the eager glue behind the LSTM chains; the
+        # column-split kernels themselves are serialized device-wide by the
+        # bindings' px event chain (two spinning px grids contend for CU
+        # residency), so overlap is safe at any batch. R2D2_NO_OVERLAP=1 off.
         overlap_ok = (self.device.type == "cuda"
-                      and cfg.learner.batch_size <= 256
                       and not _os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()

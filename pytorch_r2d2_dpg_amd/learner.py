@@ -99,11 +99,13 @@ class Learner:
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
         import os as _os
-        # Stream overlap hides the eager glue behind the LSTM chains; the
-        # column-split kernels themselves are serialized device-wide by the
-        # bindings' px event chain (two spinning px grids contend for CU
-        # residency), so overlap is safe at any batch. R2D2_NO_OVERLAP=1 off.
+        # Stream overlap pays at B <= 256 (glue hides behind the LSTM
+        # chains, +20% measured); at larger batches two concurrent
+        # column-split grids contend for CU residency and crawl, so run the
+        # chains sequentially there (the px kernels fill the chip anyway).
+        # R2D2_NO_OVERLAP=1 forces off.
         overlap_ok = (self.device.type == "cuda"
+                      and cfg.learner.batch_size <= 256
                       and not _os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()

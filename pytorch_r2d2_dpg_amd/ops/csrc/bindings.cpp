@@ -55,6 +55,8 @@ hipStream_t cur_stream() {
 // back-to-back). Chain every px launch behind the previous one with a
 // per-device event — px kernels serialize across ALL streams while the
 // surrounding GEMMs/elementwise still overlap them.
+bool g_px_serialize = false;  // default off: measured best at B<=256
+
 hipEvent_t px_chain_event(int dev) {
     static std::unordered_map<int, hipEvent_t> events;
     auto it = events.find(dev);
@@ -67,12 +69,16 @@ hipEvent_t px_chain_event(int dev) {
 }
 
 void px_chain_wait(hipStream_t st) {
-    (void)hipStreamWaitEvent(st, px_chain_event(at::cuda::current_device()), 0);
+    if (g_px_serialize)
+        (void)hipStreamWaitEvent(st, px_chain_event(at::cuda::current_device()), 0);
 }
 
 void px_chain_record(hipStream_t st) {
-    (void)hipEventRecord(px_chain_event(at::cuda::current_device()), st);
+    if (g_px_serialize)
+        (void)hipEventRecord(px_chain_event(at::cuda::current_device()), st);
 }
+
+void set_px_serialize(bool on) { g_px_serialize = on; }
 
 // Per-device error word for the column-split kernel's bounded spins.
 torch::Tensor& px_err_word(const torch::Device& dev) {
@@ -345,4 +351,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("inv_scale") = 1.0);
     m.def("lstm_px_error", &lstm_px_error,
           "read-and-clear the column-split LSTM spin-timeout error word");
+    m.def("set_px_serialize", &set_px_serialize,
+          "serialize column-split kernels device-wide (off by default)");
 }

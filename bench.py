@@ -83,6 +83,8 @@ def main() -> int:
     p.add_argument("--hidden", type=int, default=512)
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--replay-capacity", type=int, default=8192)
+    p.add_argument("--fp8-obs", action="store_true",
+                   help="store observations fp8 e4m3 in the replay shard (K7)")
     p.add_argument("--device", type=str, default="")
     p.add_argument("--graph", action="store_true", default=False,
                    help="capture the learner step in a hipGraph (default off: "
@@ -112,6 +114,7 @@ def main() -> int:
         hidden_size=cfg.model.hidden_size, per_alpha=cfg.replay.per_alpha,
         device=device, seed=cfg.learner.seed + 17 * rank,
         store_dtype=torch.bfloat16 if args.dtype == "bf16" else torch.float32,
+        fp8_obs=args.fp8_obs,
     )
     fill_replay_synthetic(replay, spec, cfg, min(2048, cfg.replay.capacity),
                           seed=cfg.learner.seed + 31 * rank)
@@ -178,7 +181,8 @@ def main() -> int:
                 "train_len": T,
                 "n_step": cfg.replay.n_step,
                 "parallelism": f"dp{n_gpus}",
-                "replay": "GPU shard, prioritized (K5/K6), per-rank",
+                "replay": "GPU shard, prioritized (K5/K6), per-rank"
+                          + (", fp8 obs (K7)" if args.fp8_obs else ""),
                 "hipgraph": graphed,
             },
         }

@@ -54,7 +54,8 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
         load_checkpoint(learner, resume)
 
     actors_per_rank = max(1, cfg.actor.num_actors // world)
-    pool = ActorPool(cfg, actors_per_rank, base_actor_id=rank * actors_per_rank)
+    pool = ActorPool(cfg, actors_per_rank, base_actor_id=rank * actors_per_rank,
+                     eval_actors=cfg.actor.eval_actors if rank == 0 else 0)
     pool.start()
     logger = MetricsLogger(log_path, stdout_every=1, rank=rank)
 
@@ -102,6 +103,8 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
                     actor_fps=(env_steps - env_steps_last) / dt,
                     replay_size=len(replay), replay_added=replay.total_added,
                     episode_return=float(np.mean(recent)) if recent else 0.0,
+                    eval_return=float(np.mean([r for _, r in pool.eval_returns[-10:]]))
+                    if pool.eval_returns else 0.0,
                     actor_respawns=pool.respawn_count,
                 )
                 t_last, trans_acc, env_steps_last = now, 0, env_steps

@@ -53,10 +53,11 @@ def unpack_sd(template: Dict[str, torch.Tensor], buf: torch.Tensor) -> Dict[str,
 
 # -------------------------------------------------------------- actor proc
 def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver,
-                stop_event, heartbeat, crash_at: int, chunk: int = 64):
+                stop_event, heartbeat, crash_at: int, chunk: int = 64,
+                eval_mode: bool = False):
     torch.set_num_threads(1)
     cfg = Config.from_dict(cfg_dict)
-    worker = ActorWorker(cfg, actor_id=actor_id)
+    worker = ActorWorker(cfg, actor_id=actor_id, eval_mode=eval_mode)
     a_tpl = worker.actor_net.state_dict()
     c_tpl = worker.critic_net.state_dict()
     na = flat_size(a_tpl)
@@ -71,6 +72,8 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
         seqs = worker.rollout(chunk)
         if crash_at and worker.env_steps >= crash_at:
             raise RuntimeError(f"fault injection: actor {actor_id} crash at {crash_at}")
+        if eval_mode:
+            seqs = []  # noise-free eval actors report episode returns only
         if seqs:
             payload = [(s.obs, s.action, s.reward, s.done, s.state, s.priority, s.length)
                        for s in seqs]
@@ -81,7 +84,8 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
         new_eps = worker.episode_returns[ep_reported:]
         if new_eps:
             try:
-                seq_queue.put(("episodes", actor_id, new_eps), timeout=0.1)
+                seq_queue.put(("eval_episodes" if eval_mode else "episodes",
+                               actor_id, new_eps), timeout=0.1)
                 ep_reported = len(worker.episode_returns)
             except queue_mod.Full:
                 pass
@@ -91,16 +95,17 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
 class ActorPool:
     def __init__(self, cfg: Config, num_actors: int, base_actor_id: int = 0,
                  crash_at: Optional[Dict[int, int]] = None, chunk: int = 64,
-                 queue_size: int = 512):
+                 queue_size: int = 512, eval_actors: int = 0):
         self.cfg = cfg
-        self.num_actors = num_actors
+        self.num_actors = num_actors + eval_actors
+        self._eval_from = num_actors  # pool indices >= this run noise-free
         self.base_actor_id = base_actor_id
         self.crash_at = crash_at or {}
         self.chunk = chunk
         self._ctx = mp.get_context("spawn")
         self.seq_queue = self._ctx.Queue(maxsize=queue_size)
         self.stop_event = self._ctx.Event()
-        self.heartbeat = self._ctx.Array("l", num_actors)
+        self.heartbeat = self._ctx.Array("l", self.num_actors)
         self.weight_ver = self._ctx.Value("l", 0)
         # shared flat weight buffer sized from a probe worker
         probe = ActorWorker(cfg, actor_id=0)
@@ -111,11 +116,12 @@ class ActorPool:
         self.weight_buf = torch.zeros(self._na + self._nc, dtype=torch.float32)
         self.weight_buf.share_memory_()
         del probe
-        self.procs: List[Optional[mp.Process]] = [None] * num_actors
-        self._last_beat = [0] * num_actors
-        self._last_beat_time = [time.time()] * num_actors
+        self.procs: List[Optional[mp.Process]] = [None] * self.num_actors
+        self._last_beat = [0] * self.num_actors
+        self._last_beat_time = [time.time()] * self.num_actors
         self.respawn_count = 0
         self.episode_returns: List[Tuple[int, float]] = []
+        self.eval_returns: List[Tuple[int, float]] = []
         self.total_sequences = 0
 
     # ------------------------------------------------------------ lifecycle
@@ -128,7 +134,8 @@ class ActorPool:
             target=_actor_main,
             args=(self.base_actor_id + i, self.cfg.to_dict(), self.seq_queue,
                   self.weight_buf, self.weight_ver, self.stop_event, self.heartbeat,
-                  self.crash_at.get(self.base_actor_id + i, 0), self.chunk),
+                  self.crash_at.get(self.base_actor_id + i, 0), self.chunk,
+                  i >= self._eval_from),
             daemon=True,
         )
         p.start()
@@ -157,6 +164,10 @@ class ActorPool:
             if isinstance(payload, tuple) and payload and payload[0] == "episodes":
                 _, aid, eps = payload
                 self.episode_returns.extend((aid, r) for r in eps)
+                continue
+            if isinstance(payload, tuple) and payload and payload[0] == "eval_episodes":
+                _, aid, eps = payload
+                self.eval_returns.extend((aid, r) for r in eps)
                 continue
             for obs, act, rew, done, state, prio, length in payload:
                 out.append(Sequence(obs, act, rew, done, state, prio, length))

@@ -207,3 +207,81 @@ def test_dp_equivalence_vs_single(tmp_path):
         p.join(timeout=30)
     np.testing.assert_allclose(results[0], want, atol=2e-5)
     np.testing.assert_allclose(results[1], want, atol=2e-5)
+
+
+def _ramp_worker(rank, world, port, out_q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from pytorch_r2d2_dpg_amd.parallel.dist import DistContext
+
+    dctx = DistContext.from_env(backend="gloo")
+    # ramp pattern: grads = ramp * (rank+1); mean = ramp * 1.5
+    params = [torch.nn.Parameter(torch.zeros(n)) for n in (17, 64, 5)]
+    off = 0
+    for p in params:
+        p.grad = torch.arange(off, off + p.numel(), dtype=torch.float32) * (rank + 1)
+        off += p.numel()
+    dctx.allreduce_grads(params, key="ramp")
+    flat = torch.cat([p.grad for p in params])
+    out_q.put((rank, flat.numpy()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_allreduce_ramp_pattern():
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ramp_worker, args=(r, 2, 29581, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        r, f = q.get(timeout=90)
+        res[r] = f
+    for p in procs:
+        p.join(timeout=20)
+    want = np.arange(17 + 64 + 5, dtype=np.float32) * 1.5
+    np.testing.assert_allclose(res[0], want, rtol=1e-6)
+    np.testing.assert_allclose(res[1], want, rtol=1e-6)
+
+
+def test_replay_shard_isolation():
+    """Per-rank shards with different seeds sample independently."""
+    from pytorch_r2d2_dpg_amd.replay.gpu_buffer import GpuSequenceReplay
+    import torch
+
+    def mk(seed):
+        rep = GpuSequenceReplay(capacity=64, stored_len=6, obs_shape=(3,), action_dim=2,
+                                hidden_size=8, per_alpha=1.0, device="cpu", seed=seed)
+        g = torch.Generator().manual_seed(0)
+        rep.add_batch(torch.randn(64, 6, 3, generator=g), torch.randn(64, 6, 2, generator=g),
+                      torch.rand(64, 6, generator=g), torch.zeros(64, 6),
+                      torch.zeros(64, 4, 8), torch.rand(64, generator=g) + 0.1)
+        return rep
+
+    a, b = mk(1), mk(2)
+    ia = a.sample(32)["idx"]
+    ib = b.sample(32)["idx"]
+    assert not torch.equal(ia, ib), "different shard seeds must sample differently"
+    # same seed => identical sampling (per-rank determinism)
+    c = mk(1)
+    ic = c.sample(32)["idx"]
+    assert torch.equal(ia, ic)
+
+
+@pytest.mark.timeout(120)
+def test_eval_actors_report_noise_free_returns():
+    from pytorch_r2d2_dpg_amd.parallel.actor_pool import ActorPool
+    cfg = pool_cfg()
+    pool = ActorPool(cfg, num_actors=1, eval_actors=1, chunk=30)
+    pool.start()
+    try:
+        deadline = time.time() + 60
+        while not pool.eval_returns and time.time() < deadline:
+            pool.poll()
+            time.sleep(0.2)
+        assert pool.eval_returns, "eval actor reported no episodes"
+    finally:
+        pool.stop()

@@ -154,6 +154,14 @@ def main() -> int:
     dist.barrier()
     elapsed = dist.allreduce_scalar(elapsed, op="max")
 
+    # safety net: the column-split kernels flag bounded-spin timeouts here
+    if device.startswith("cuda"):
+        from pytorch_r2d2_dpg_amd.ops import hip_ext
+        px_err = hip_ext().lstm_px_error()
+        if px_err:
+            print(f"[bench] WARNING: px spin-timeout flagged (code {px_err}) — "
+                  "numerics of some steps are unreliable", file=sys.stderr, flush=True)
+
     ms_per_step = elapsed / args.steps * 1000.0
     n_gpus = world if world > 1 else args.gpus
     value = n_gpus * B * T / (elapsed / args.steps)

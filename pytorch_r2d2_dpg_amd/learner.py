@@ -72,9 +72,19 @@ class Learner:
                 t.copy_(o)
 
         lc = cfg.learner
-        self.opt_actor = torch.optim.Adam(self.actor.parameters(), lr=lc.lr_actor, eps=lc.adam_eps)
-        self.opt_critic = torch.optim.Adam(self.critic.parameters(), lr=lc.lr_critic, eps=lc.adam_eps,
-                                           weight_decay=lc.critic_weight_decay)
+        fused_ok = self.device.type == "cuda"
+        try:
+            self.opt_actor = torch.optim.Adam(self.actor.parameters(), lr=lc.lr_actor,
+                                              eps=lc.adam_eps, fused=fused_ok)
+            self.opt_critic = torch.optim.Adam(self.critic.parameters(), lr=lc.lr_critic,
+                                               eps=lc.adam_eps, fused=fused_ok,
+                                               weight_decay=lc.critic_weight_decay)
+        except (RuntimeError, ValueError):  # fused unsupported on this build
+            self.opt_actor = torch.optim.Adam(self.actor.parameters(), lr=lc.lr_actor,
+                                              eps=lc.adam_eps)
+            self.opt_critic = torch.optim.Adam(self.critic.parameters(), lr=lc.lr_critic,
+                                               eps=lc.adam_eps,
+                                               weight_decay=lc.critic_weight_decay)
         self.polyak_actor = PolyakUpdater(
             list(self.target_actor.parameters()), list(self.actor.parameters()))
         self.polyak_critic = PolyakUpdater(

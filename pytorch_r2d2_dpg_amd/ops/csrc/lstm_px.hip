@@ -89,6 +89,18 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     const long slot_sz = (long)gridDim.x * 16 * (H / 2);
 
     for (int t = 0; t < T; ++t) {
+        const __hip_bfloat16* gx_t = gates_x + (long)t * B * H4;
+        // gates_x accumulator init first: these HBM loads are independent of
+        // the exchange, so they overlap the granule sweep below
+        f32x4 acc[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int grow = brow0 + c_row0 + r;
+                int j = g * H + col_base + c_col;
+                acc[g][r] = (grow < B) ? bf2f(gx_t[(long)grow * H4 + j]) : 0.0f;
+            }
         // ---- sweep ALL h(t-1) granules (peers' and own) into LDS buf[t&1]
         // (skip t=0: h0 already local). Batched: each thread issues its 2*CG
         // granule loads unconditionally (one L2 round-trip covers all),
@@ -140,21 +152,11 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
         }
         __syncthreads();
 
-        const __hip_bfloat16* gx_t = gates_x + (long)t * B * H4;
         __hip_bfloat16* hs_t = h_seq + (long)t * B * H;
         __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
         float* cs_t = c_seq + (long)t * B * H;
         char* rd_buf = smem + (t & 1) * tile_bytes;
 
-        f32x4 acc[4];
-#pragma unroll
-        for (int g = 0; g < 4; ++g)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int grow = brow0 + c_row0 + r;
-                int j = g * H + col_base + c_col;
-                acc[g][r] = (grow < B) ? bf2f(gx_t[(long)grow * H4 + j]) : 0.0f;
-            }
         for (int kk = 0; kk < H; kk += 32) {
             int k = kk + a_kgrp;
             short8 a_frag = *reinterpret_cast<const short8*>(

@@ -11,7 +11,7 @@ process (parallel/actor_pool.py).
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 import numpy as np
 import torch

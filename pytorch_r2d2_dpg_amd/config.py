@@ -15,7 +15,6 @@ from __future__ import annotations
 import dataclasses
 import json
 from dataclasses import dataclass, field
-from typing import Optional
 
 try:  # py3.11+ stdlib, else the wheelhouse 'tomli'
     import tomllib as _toml

@@ -33,9 +33,7 @@ from .config import Config
 from .models import make_models
 from .ops.polyak import PolyakUpdater
 from .ops.td import td_targets_and_priorities
-from .ops.reference import value_rescale, value_rescale_inv
 from .parallel.dist import DistContext
-from .replay.buffer import SequenceReplay
 
 
 def _null_ctx():

@@ -12,7 +12,7 @@ Target copies are deep copies updated by Polyak (ops/polyak.py).
 from __future__ import annotations
 
 import copy
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 from torch import nn

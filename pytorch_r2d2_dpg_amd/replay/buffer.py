@@ -15,7 +15,7 @@ sampled batch gather never leaves HBM.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence as Seq
+from typing import Dict, Sequence as Seq
 
 import numpy as np
 import torch

@@ -15,7 +15,7 @@ especially); decode happens inside the fused gather.
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 import numpy as np
 import torch

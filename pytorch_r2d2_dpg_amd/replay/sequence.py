@@ -17,7 +17,7 @@ zero-reward terminal steps, mask excluded via done flags) and reset.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 

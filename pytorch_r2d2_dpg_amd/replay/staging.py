@@ -10,7 +10,7 @@ publish-last ordering (§3.4) without ever blocking on the copy.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 import torch

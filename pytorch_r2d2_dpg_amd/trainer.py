@@ -10,8 +10,6 @@ from __future__ import annotations
 import time
 from typing import Optional
 
-import torch
-
 from .actor import ActorWorker
 from .checkpoint import save_checkpoint
 from .config import Config

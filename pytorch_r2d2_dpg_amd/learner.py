@@ -32,6 +32,7 @@ import torch
 from .config import Config
 from .models import make_models
 from .ops.polyak import PolyakUpdater
+from .ops.lstm import lstm_seq_pair
 from .ops.td import td_targets_and_priorities
 from .parallel.dist import DistContext
 
@@ -184,22 +185,24 @@ class Learner:
             s1.wait_stream(main)
 
         self._pe("burn_in", 0)
-        # ---- burn-ins: actor chain on main, critic chain on the side stream
+        # ---- burn-in: BOTH no-grad unrolls in ONE paired column-split
+        # launch (grid.z = 2) — a single grid co-dispatches uniformly, so
+        # the two halves genuinely parallelize (two px launches on separate
+        # streams contend for CU residency instead)
         if bi > 0:
             with torch.no_grad():
-                _, (h_a2, c_a2) = self.actor(obs[:bi], (h_a, c_a))
+                feat_a = self.actor.trunk(obs[:bi])
+                feat_c = torch.cat([self.critic.trunk(obs[:bi]), action[:bi]], dim=-1)
+                la, lc = self.actor.lstm, self.critic.lstm
+                _, h_a2, c_a2, _, h_c2, c_c2 = lstm_seq_pair(
+                    feat_a, h_a, c_a, la.weight_x, la.weight_h, la.bias,
+                    feat_c, h_c, c_c, lc.weight_x, lc.weight_h, lc.bias)
         else:
-            h_a2, c_a2 = h_a, c_a
+            h_a2, c_a2, h_c2, c_c2 = h_a, c_a, h_c, c_c
         if overlap:
             ev_ba.record(main)
         with torch.cuda.stream(s1) if overlap else _null_ctx():
-            if bi > 0:
-                with torch.no_grad():
-                    _, (h_c2, c_c2) = self.critic(obs[:bi], action[:bi], (h_c, c_c))
-            else:
-                h_c2, c_c2 = h_c, c_c
             if overlap:
-                ev_bc.record(s1)
                 s1.wait_event(ev_ba)
             # actor chain through the PRE-update critic
             mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
@@ -215,8 +218,6 @@ class Learner:
         # side stream's actor-chain length)
         with torch.no_grad():
             mu_t, _ = self.target_actor(obs[bi:bi + Tn], (h_a2, c_a2))
-            if overlap:
-                main.wait_event(ev_bc)
             qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
             q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
         q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],

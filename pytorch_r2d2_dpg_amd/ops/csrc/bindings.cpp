@@ -39,6 +39,11 @@ extern "C" void launch_fp8_gather_decode(const unsigned char*, const long*, void
 extern "C" int launch_lstm_fwd_px_bf16(const void*, const void*, const void*, const void*,
                                        void*, void*, void*, float*, unsigned long long*,
                                        unsigned int*, int, int, int, int, hipStream_t);
+extern "C" int launch_lstm_fwd_px2_bf16(const void*, const void*, const void*, const void*,
+                                        void*, void*, unsigned long long*,
+                                        const void*, const void*, const void*, const void*,
+                                        void*, void*, unsigned long long*,
+                                        unsigned int*, int, int, int, hipStream_t);
 extern "C" int launch_lstm_bwd_px_bf16(const void*, const void*, const void*, const void*,
                                        const float*, const void*, const void*, void*, void*,
                                        void*, unsigned char*, unsigned long long*,
@@ -178,6 +183,39 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
                                  (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
     TORCH_CHECK(rc == 0, "unsupported LSTM geometry H=", H);
     return {h_seq, c_final, gates_act, c_seq};
+}
+
+// ---- K1-pair: two independent no-grad LSTM unrolls in ONE launch ----------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
+                  torch::Tensor wh0, torch::Tensor gx1, torch::Tensor h01,
+                  torch::Tensor c01, torch::Tensor wh1) {
+    for (auto* t : {&gx0, &h00, &c00, &wh0, &gx1, &h01, &c01, &wh1})
+        check_dev_contig(*t, "pair arg");
+    const auto T = gx0.size(0);
+    const auto B = gx0.size(1);
+    const auto H = h00.size(1);
+    TORCH_CHECK(gx1.size(0) == T && gx1.size(1) == B && h01.size(1) == H,
+                "pair shape mismatch");
+    TORCH_CHECK(gx0.scalar_type() == torch::kBFloat16, "pair path is bf16-only");
+    auto wp0 = pack_fragment_major(wh0);
+    auto wp1 = pack_fragment_major(wh1);
+    const int RT = (int)((B + 15) / 16);
+    auto hx0 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
+    auto hx1 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
+    auto hs0 = torch::empty({T, B, H}, gx0.options());
+    auto hs1 = torch::empty({T, B, H}, gx0.options());
+    auto cf0 = torch::empty({B, H}, gx0.options());
+    auto cf1 = torch::empty({B, H}, gx0.options());
+    auto& err = px_err_word(gx0.device());
+    int rc = launch_lstm_fwd_px2_bf16(
+        gx0.data_ptr(), h00.data_ptr(), c00.data_ptr(), wp0.data_ptr(),
+        hs0.data_ptr(), cf0.data_ptr(), (unsigned long long*)hx0.data_ptr<int64_t>(),
+        gx1.data_ptr(), h01.data_ptr(), c01.data_ptr(), wp1.data_ptr(),
+        hs1.data_ptr(), cf1.data_ptr(), (unsigned long long*)hx1.data_ptr<int64_t>(),
+        (unsigned int*)err.data_ptr<int32_t>(), (int)T, (int)B, (int)H, cur_stream());
+    TORCH_CHECK(rc == 0, "pair geometry unsupported (fall back to two calls)");
+    return {hs0, cf0, hs1, cf1};
 }
 
 // ---- K2: fused LSTM sequence backward -------------------------------------
@@ -351,6 +389,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("inv_scale") = 1.0);
     m.def("lstm_px_error", &lstm_px_error,
           "read-and-clear the column-split LSTM spin-timeout error word");
+    m.def("lstm_seq_fwd_pair", &lstm_seq_fwd_pair,
+          "two independent no-grad LSTM unrolls in one launch");
     m.def("set_px_serialize", &set_px_serialize,
           "serialize column-split kernels device-wide (off by default)");
 }

@@ -512,3 +512,205 @@ extern "C" int launch_lstm_bwd_px_bf16(
 #undef BWD_PX_CALL
     return 0;
 }
+
+// ------------------------------------------------- paired (z=2) forward
+// Two INDEPENDENT same-shape LSTMs (e.g. the actor and critic burn-in
+// unrolls) in one launch: grid.z selects the pointer set. A single grid
+// dispatches uniformly, so the two halves co-reside by construction —
+// unlike two concurrent px launches on different streams, which contend
+// for CUs and run ~3x slower than back-to-back.
+struct PxArgs {
+    const __hip_bfloat16* gates_x;
+    const __hip_bfloat16* h0;
+    const __hip_bfloat16* c0;
+    const __hip_bfloat16* wh;
+    __hip_bfloat16* h_seq;
+    __hip_bfloat16* c_final;
+    unsigned long long* hx;
+};
+
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int T, int B, int H) {
+    const PxArgs& a = blockIdx.z ? a1 : a0;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int rt = blockIdx.x;
+    const int cg = blockIdx.y;
+    const int brow0 = rt * 16;
+    const int HQg = H / CG;
+    const int col_base = cg * HQg + wid * 16;
+    const int H4 = 4 * H;
+    const int rowbytes = H * 2;
+    const unsigned tile_bytes = 16u * (unsigned)rowbytes;
+
+    const int a_row = lane & 15;
+    const int a_kgrp = (lane >> 4) * 8;
+    const int c_col = lane & 15;
+    const int c_row0 = (lane >> 4) * 4;
+
+    for (int e = tid; e < 16 * H; e += (int)blockDim.x) {
+        int r = e / H, col = e - r * H;
+        int grow = brow0 + r;
+        float v = (grow < B) ? bf2f(a.h0[(long)grow * H + col]) : 0.0f;
+        *reinterpret_cast<__hip_bfloat16*>(&smem[px_swz(r * rowbytes + col * 2, r)]) = f2bf(v);
+    }
+    float c_reg[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        c_reg[r] = (grow < B) ? bf2f(a.c0[(long)grow * H + col]) : 0.0f;
+    }
+    __syncthreads();
+
+    gu64* hx_g = (gu64*)a.hx;
+    const long rt_off = (long)rt * 16 * (H / 2);
+    const long slot_sz = (long)gridDim.x * 16 * (H / 2);
+
+    for (int t = 0; t < T; ++t) {
+        const __hip_bfloat16* gx_t = a.gates_x + (long)t * B * H4;
+        f32x4 acc[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int grow = brow0 + c_row0 + r;
+                int j = g * H + col_base + c_col;
+                acc[g][r] = (grow < B) ? bf2f(gx_t[(long)grow * H4 + j]) : 0.0f;
+            }
+        if (t > 0) {
+            char* bld = smem + (t & 1) * tile_bytes;
+            const unsigned tag = (unsigned)t;
+            gu64* slot = hx_g + ((t - 1) & 1) * slot_sz + rt_off;
+            constexpr int K = 2 * CG;
+            const int bd = (int)blockDim.x;
+            const int sh = 31 - __builtin_clz((unsigned)(H >> 1));
+            unsigned long long v[K];
+#pragma unroll
+            for (int i = 0; i < K; ++i)
+                v[i] = __hip_atomic_load(&slot[tid + i * bd], RLX_AGENT);
+            unsigned pending = 0;
+#pragma unroll
+            for (int i = 0; i < K; ++i)
+                pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
+            unsigned spins = 0;
+            while (pending) {
+                if (++spins > 200000000u) {  // convergent bail (see single fwd)
+                    __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
+                    break;
+                }
+                __builtin_amdgcn_s_sleep(1);
+#pragma unroll
+                for (int i = 0; i < K; ++i)
+                    if (pending & (1u << i)) {
+                        unsigned long long nv =
+                            __hip_atomic_load(&slot[tid + i * bd], RLX_AGENT);
+                        if ((unsigned)(nv >> 32) == tag) {
+                            v[i] = nv;
+                            pending &= ~(1u << i);
+                        }
+                    }
+            }
+#pragma unroll
+            for (int i = 0; i < K; ++i) {
+                int e = tid + i * bd;
+                int r = e >> sh, gcol = e & ((H >> 1) - 1);
+                *reinterpret_cast<unsigned*>(
+                    &bld[px_swz(r * rowbytes + gcol * 4, r)]) = (unsigned)v[i];
+            }
+        }
+        __syncthreads();
+
+        __hip_bfloat16* hs_t = a.h_seq + (long)t * B * H;
+        char* rd_buf = smem + (t & 1) * tile_bytes;
+        for (int kk = 0; kk < H; kk += 32) {
+            int k = kk + a_kgrp;
+            short8 a_frag = *reinterpret_cast<const short8*>(
+                &rd_buf[px_swz(a_row * rowbytes + k * 2, a_row)]);
+            short8 breg[4];
+            const int kb = kk >> 5;
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                int jb = g * (H >> 4) + (col_base >> 4);
+                long off = (((long)jb * (H >> 5) + kb) * 64 + lane) * 8;
+                breg[g] = *reinterpret_cast<const short8*>(&a.wh[off]);
+            }
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+                acc[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, breg[g], acc[g], 0, 0, 0);
+        }
+        {
+            gu64* slot = hx_g + (t & 1) * slot_sz + rt_off;
+            const unsigned tag = (unsigned)(t + 1);
+            const int col = col_base + c_col;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = c_row0 + r;
+                int grow = brow0 + row;
+                float gi = px_sigm(acc[0][r]);
+                float gf = px_sigm(acc[1][r]);
+                float gg = px_tanh(acc[2][r]);
+                float go = px_sigm(acc[3][r]);
+                float c = gf * c_reg[r] + gi * gg;
+                float hn = go * px_tanh(c);
+                c_reg[r] = c;
+                unsigned short hb = f32_to_bf16_bits(hn);
+                unsigned other = __shfl_xor((unsigned)hb, 1);
+                if ((lane & 1) == 0) {
+                    unsigned lo = (unsigned)hb | (other << 16);
+                    __hip_atomic_store(&slot[row * (H / 2) + col / 2],
+                                       ((unsigned long long)tag << 32) | lo, RLX_AGENT);
+                }
+                if (grow < B) hs_t[(long)grow * H + col] = f2bf(hn);
+            }
+        }
+        __syncthreads();
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        if (grow < B) a.c_final[(long)grow * H + col] = f2bf(c_reg[r]);
+    }
+}
+
+extern "C" int launch_lstm_fwd_px2_bf16(
+    const void* gx0, const void* h00, const void* c00, const void* wh0,
+    void* hs0, void* cf0, unsigned long long* hx0,
+    const void* gx1, const void* h01, const void* c01, const void* wh1,
+    void* hs1, void* cf1, unsigned long long* hx1,
+    unsigned int* err, int T, int B, int H, hipStream_t stream) {
+    int RT = (B + 15) / 16;
+    if (H & (H - 1)) return -1;
+    int CG = 0;
+    for (int c : {8, 4, 2}) {
+        if (H % (16 * c)) continue;
+        int w = (H / c) / 16;
+        if (w < 2 || w > 16) continue;
+        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
+        CG = c;
+        break;
+    }
+    if (CG == 0) return -1;
+    int waves = (H / CG) / 16;
+    dim3 grid(RT, CG, 2);
+    size_t lds = (size_t)2 * 16 * H * 2;
+    PxArgs a0{(const __hip_bfloat16*)gx0, (const __hip_bfloat16*)h00,
+              (const __hip_bfloat16*)c00, (const __hip_bfloat16*)wh0,
+              (__hip_bfloat16*)hs0, (__hip_bfloat16*)cf0, hx0};
+    PxArgs a1{(const __hip_bfloat16*)gx1, (const __hip_bfloat16*)h01,
+              (const __hip_bfloat16*)c01, (const __hip_bfloat16*)wh1,
+              (__hip_bfloat16*)hs1, (__hip_bfloat16*)cf1, hx1};
+    switch (CG) {
+        case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
+        case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
+        case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
+        default: return -1;
+    }
+    return 0;
+}

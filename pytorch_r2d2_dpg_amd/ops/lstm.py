@@ -95,3 +95,25 @@ def lstm_seq(
         h_seq, cT = _FusedLSTMSeq.apply(x, h0, c0, wx, wh, b)
         return h_seq, h_seq[-1], cT
     return lstm_seq_ref(x, h0, c0, wx, wh, b)
+
+
+def lstm_seq_pair(x0, h00, c00, wx0, wh0, b0, x1, h01, c01, wx1, wh1, b1):
+    """Two independent no-grad LSTM unrolls (same T, B, H) in ONE launch
+    (burn-in actor + critic). Falls back to two sequential calls when the
+    paired geometry is unsupported. Returns (h_seq0, hT0, cT0, h_seq1, hT1, cT1)."""
+    if x0.is_cuda and x0.dtype == torch.bfloat16 and not torch.is_grad_enabled():
+        T, B, _ = x0.shape
+        H = h00.shape[1]
+        ext = hip_ext()
+        gx0 = torch.addmm(b0, x0.reshape(T * B, -1), wx0.t()).view(T, B, 4 * H)
+        gx1 = torch.addmm(b1, x1.reshape(T * B, -1), wx1.t()).view(T, B, 4 * H)
+        try:
+            hs0, cf0, hs1, cf1 = ext.lstm_seq_fwd_pair(
+                gx0.contiguous(), h00.contiguous(), c00.contiguous(), wh0.contiguous(),
+                gx1.contiguous(), h01.contiguous(), c01.contiguous(), wh1.contiguous())
+            return hs0, hs0[-1], cf0, hs1, hs1[-1], cf1
+        except RuntimeError:
+            pass
+    a = lstm_seq(x0, h00, c00, wx0, wh0, b0)
+    c = lstm_seq(x1, h01, c01, wx1, wh1, b1)
+    return a[0], a[1], a[2], c[0], c[1], c[2]

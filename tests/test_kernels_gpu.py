@@ -352,3 +352,27 @@ def test_lstm_bwd_px_vs_ref(shape):
         num = (rc_.grad - gc.grad.float().cpu()).norm()
         den = rc_.grad.norm().clamp_min(1e-3)
         assert num / den < 0.09, f"d{name} rel-norm err {num / den}"
+
+
+def test_lstm_fwd_pair_matches_singles():
+    """Paired (grid.z=2) burn-in launch must match two single launches."""
+    T, B, I, H = 9, 64, 40, 512
+    g = torch.Generator().manual_seed(31)
+    def mk():
+        x = torch.randn(T, B, I, generator=g).cuda().bfloat16()
+        h0 = (torch.randn(B, H, generator=g) * 0.3).cuda().bfloat16()
+        c0 = (torch.randn(B, H, generator=g) * 0.3).cuda().bfloat16()
+        wx = (torch.randn(4 * H, I, generator=g) / I ** 0.5).cuda().bfloat16()
+        wh = (torch.randn(4 * H, H, generator=g) / H ** 0.5).cuda().bfloat16()
+        b = (torch.randn(4 * H, generator=g) * 0.1).cuda().bfloat16()
+        return x, h0, c0, wx, wh, b
+    a, c = mk(), mk()
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq, lstm_seq_pair
+    with torch.no_grad():
+        hs0, hT0, cT0, hs1, hT1, cT1 = lstm_seq_pair(*a, *c)
+        r0 = lstm_seq(*a)
+        r1 = lstm_seq(*c)
+    torch.cuda.synchronize()
+    assert ext().lstm_px_error() == 0
+    assert torch.equal(hs0, r0[0]) and torch.equal(cT0, r0[2])
+    assert torch.equal(hs1, r1[0]) and torch.equal(cT1, r1[2])

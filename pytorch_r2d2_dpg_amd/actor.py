@@ -62,6 +62,9 @@ class ActorWorker:
     def rollout(self, num_steps: int) -> List[Sequence]:
         """Run `num_steps` env steps; return the sequences cut along the way."""
         out: List[Sequence] = []
+        cap = self.cfg.actor.max_env_steps
+        if cap and self.env_steps >= cap:
+            return out
         h_a, c_a, h_c, c_c = self._state
         for _ in range(num_steps):
             snapshot = np.stack([h_a[0].numpy(), c_a[0].numpy(),
@@ -78,7 +81,15 @@ class ActorWorker:
                             -self.spec.action_scale, self.spec.action_scale)
             q, (h_c, c_c) = self.critic_net.step(
                 a_feat, torch.from_numpy(a).unsqueeze(0), (h_c, c_c))
-            next_obs, r, done, trunc = self.env.step(a)
+            # action repeat (env wrapper semantic): apply a for k frames,
+            # accumulate reward, stop early on termination
+            r = 0.0
+            done = trunc = False
+            for _ in range(max(self.cfg.env.action_repeat, 1)):
+                next_obs, rk, done, trunc = self.env.step(a)
+                r += rk
+                if done or trunc:
+                    break
             self.episode_return += r
             self.env_steps += 1
             out.extend(self.cutter.push(self._obs, a, r * self.cfg.replay.reward_scale,

@@ -53,7 +53,6 @@ class ReplayConfig:
     min_fill: int = 256             # sequences before learning starts
     reward_scale: float = 1.0       # rewards are scaled at actor-push time
     fp8_obs: bool = False           # store observations as fp8 e4m3 in the GPU shard
-    device: str = "auto"            # auto: GPU shard when learner is on GPU
 
 
 @dataclass

@@ -36,3 +36,45 @@ def test_synthetic_pixel_uint8():
     env = make_env("quadruped-run-pixels")
     obs = env.reset()
     assert obs.dtype == np.uint8 and obs.shape == (84, 84, 3)
+
+
+def test_action_repeat_accumulates_reward():
+    from pytorch_r2d2_dpg_amd.actor import ActorWorker
+    from pytorch_r2d2_dpg_amd.config import Config
+
+    cfg = Config()
+    cfg.env.name = "pendulum"
+    cfg.env.time_limit = 40
+    cfg.env.action_repeat = 4
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.replay.seq_len = 8
+    cfg.replay.burn_in = 2
+    cfg.replay.overlap = 4
+    cfg.replay.n_step = 2
+    w = ActorWorker(cfg, actor_id=0, sigma=0.2)
+    w.rollout(10)  # 10 policy steps = 40 env frames = one episode
+    assert w.env_steps == 10
+    assert len(w.episode_returns) >= 1  # time limit hit within repeats
+
+
+def test_max_env_steps_caps_rollout():
+    from pytorch_r2d2_dpg_amd.actor import ActorWorker
+    from pytorch_r2d2_dpg_amd.config import Config
+
+    cfg = Config()
+    cfg.env.name = "pendulum"
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.replay.seq_len = 8
+    cfg.replay.burn_in = 2
+    cfg.replay.overlap = 4
+    cfg.replay.n_step = 2
+    cfg.actor.max_env_steps = 15
+    w = ActorWorker(cfg, actor_id=0, sigma=0.2)
+    w.rollout(30)
+    assert w.env_steps <= 30
+    w.rollout(30)
+    more = w.env_steps
+    w.rollout(30)
+    assert w.env_steps == more  # capped: no further progress

@@ -173,3 +173,47 @@ def test_pixel_actor_rollout_and_replay():
     ring.ingest(seqs)
     b = rep.sample(4)
     assert b["obs"].dtype == torch.uint8
+
+
+def test_all_shipped_configs_parse_and_build():
+    """The five BASELINE.json configs must parse and build their models."""
+    import glob
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.models import make_models
+
+    cfgs = sorted(glob.glob(os.path.join(os.path.dirname(__file__), "..",
+                                         "configs", "*.toml")))
+    assert len(cfgs) == 5
+    for path in cfgs:
+        cfg = Config.from_toml(path)
+        spec = make_env(cfg.env.name).spec
+        a, c, ta, tc = make_models(spec, cfg.model)
+        assert sum(p.numel() for p in a.parameters()) > 0
+        assert cfg.replay.seq_len > cfg.replay.burn_in
+
+
+def test_orchestrator_resume_from_checkpoint(tmp_path):
+    from pytorch_r2d2_dpg_amd.orchestrator import run_training
+
+    cfg = Config()
+    cfg.env.name = "pendulum"
+    cfg.env.time_limit = 60
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.replay.seq_len = 12
+    cfg.replay.burn_in = 4
+    cfg.replay.overlap = 6
+    cfg.replay.n_step = 3
+    cfg.replay.capacity = 256
+    cfg.replay.min_fill = 8
+    cfg.learner.batch_size = 8
+    cfg.learner.sync_every = 5
+    cfg.learner.checkpoint_every = 6
+    cfg.learner.checkpoint_dir = str(tmp_path)
+    cfg.actor.num_actors = 1
+    lrn = run_training(cfg, device="cpu", max_steps=6)
+    ckpt = os.path.join(str(tmp_path), "ckpt_6.pth")
+    assert os.path.exists(ckpt)
+    cfg.learner.checkpoint_every = 0
+    lrn2 = run_training(cfg, device="cpu", resume=ckpt, max_steps=9)
+    assert lrn2.step_count == 9  # resumed from 6, ran to 9

@@ -21,7 +21,7 @@ import numpy as np
 import torch
 import torch.multiprocessing as mp
 
-from ..actor import ActorWorker
+from ..actor import ActorWorker, VectorActorWorker
 from ..config import Config
 from ..replay.sequence import Sequence
 
@@ -57,7 +57,11 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
                 eval_mode: bool = False):
     torch.set_num_threads(1)
     cfg = Config.from_dict(cfg_dict)
-    worker = ActorWorker(cfg, actor_id=actor_id, eval_mode=eval_mode)
+    E = max(cfg.actor.envs_per_actor, 1)
+    if E > 1 and not eval_mode:
+        worker = VectorActorWorker(cfg, actor_id=actor_id, num_envs=E)
+    else:
+        worker = ActorWorker(cfg, actor_id=actor_id, eval_mode=eval_mode)
     a_tpl = worker.actor_net.state_dict()
     c_tpl = worker.critic_net.state_dict()
     na = flat_size(a_tpl)

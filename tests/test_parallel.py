@@ -285,3 +285,17 @@ def test_eval_actors_report_noise_free_returns():
         assert pool.eval_returns, "eval actor reported no episodes"
     finally:
         pool.stop()
+
+
+@pytest.mark.timeout(120)
+def test_vector_actor_matches_interface():
+    from pytorch_r2d2_dpg_amd.actor import VectorActorWorker
+    cfg = pool_cfg()
+    w = VectorActorWorker(cfg, actor_id=0, num_envs=4)
+    seqs = w.rollout(20)  # 80 env frames
+    assert w.env_steps == 80
+    assert seqs, "vector actor produced no sequences"
+    s = seqs[0]
+    assert s.obs.shape == (cfg.replay.seq_len + cfg.replay.n_step, 3)
+    assert s.state.shape == (4, 16)
+    assert np.isfinite(s.priority)

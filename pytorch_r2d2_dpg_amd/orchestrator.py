@@ -10,6 +10,7 @@ shared-memory buffer every sync_every steps; rank 0 checkpoints and logs.
 from __future__ import annotations
 
 import os
+import threading
 import time
 from typing import Optional
 
@@ -61,17 +62,30 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
 
     staging = StagingRing(replay)
 
-    def ingest():
-        seqs = pool.poll()
-        if seqs:
-            staging.ingest(seqs)
-        return len(seqs)
+    # ---- background ingest thread: drains stacked-array payloads and
+    # stages them into the shard off the learner's critical path. Storage
+    # writes + tree publish enqueue on the learner's stream, so sampling
+    # (same stream) never sees a torn slot (§3.4 single-writer contract:
+    # this thread is the only writer).
+    stop_ingest = threading.Event()
+
+    def ingest_loop():
+        if replay.is_cuda:
+            torch.cuda.set_device(replay.device)
+        while not stop_ingest.is_set():
+            batches = pool.poll_arrays()
+            for b in batches:
+                staging.ingest_arrays(*b)
+            if not batches:
+                time.sleep(0.002)
+
+    ingest_thread = threading.Thread(target=ingest_loop, daemon=True)
+    ingest_thread.start()
 
     steps = max_steps if max_steps is not None else cfg.learner.steps
     try:
         # ---- warm the replay
         while len(replay) < max(R.min_fill, cfg.learner.batch_size):
-            ingest()
             pool.watchdog()
             time.sleep(0.05)
 
@@ -79,7 +93,6 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
         trans_acc = 0
         env_steps_last = pool.aggregate_env_steps()
         for step in range(learner.step_count, steps):
-            ingest()
             batch = replay.sample(cfg.learner.batch_size)
             stats = learner.train_step(batch)
             trans_acc += stats.transitions
@@ -109,6 +122,8 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
                 )
                 t_last, trans_acc, env_steps_last = now, 0, env_steps
     finally:
+        stop_ingest.set()
+        ingest_thread.join(timeout=5)
         pool.stop()
         logger.close()
     return learner

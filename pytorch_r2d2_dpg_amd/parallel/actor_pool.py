@@ -79,8 +79,16 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
         if eval_mode:
             seqs = []  # noise-free eval actors report episode returns only
         if seqs:
-            payload = [(s.obs, s.action, s.reward, s.done, s.state, s.priority, s.length)
-                       for s in seqs]
+            # stacked-array payload: ONE pickle buffer per field instead of
+            # per-sequence tuples (unpickle cost on the learner side was the
+            # ingest bottleneck at high actor FPS)
+            payload = ("seqs",
+                       np.stack([s.obs for s in seqs]),
+                       np.stack([s.action for s in seqs]),
+                       np.stack([s.reward for s in seqs]),
+                       np.stack([s.done for s in seqs]),
+                       np.stack([s.state for s in seqs]),
+                       np.asarray([s.priority for s in seqs], dtype=np.float32))
             try:
                 seq_queue.put(payload, timeout=5.0)
             except queue_mod.Full:
@@ -157,25 +165,35 @@ class ActorPool:
                     p.terminate()
 
     # ------------------------------------------------------------- data i/o
-    def poll(self, max_batches: int = 64) -> List[Sequence]:
-        """Drain available sequences (non-blocking)."""
-        out: List[Sequence] = []
+    def poll_arrays(self, max_batches: int = 64) -> List[tuple]:
+        """Drain stacked sequence batches (obs, act, rew, done, state, prio)
+        as numpy arrays — the fast ingest path (no per-sequence objects)."""
+        out: List[tuple] = []
         for _ in range(max_batches):
             try:
                 payload = self.seq_queue.get_nowait()
             except queue_mod.Empty:
                 break
-            if isinstance(payload, tuple) and payload and payload[0] == "episodes":
+            tag = payload[0]
+            if tag == "episodes":
                 _, aid, eps = payload
                 self.episode_returns.extend((aid, r) for r in eps)
-                continue
-            if isinstance(payload, tuple) and payload and payload[0] == "eval_episodes":
+            elif tag == "eval_episodes":
                 _, aid, eps = payload
                 self.eval_returns.extend((aid, r) for r in eps)
-                continue
-            for obs, act, rew, done, state, prio, length in payload:
-                out.append(Sequence(obs, act, rew, done, state, prio, length))
-        self.total_sequences += len(out)
+            else:
+                out.append(payload[1:])
+                self.total_sequences += payload[1].shape[0]
+        return out
+
+    def poll(self, max_batches: int = 64) -> List[Sequence]:
+        """Drain available sequences as objects (tests / small runs)."""
+        out: List[Sequence] = []
+        for batch in self.poll_arrays(max_batches):
+            obs, act, rew, done, state, prio = batch
+            for i in range(obs.shape[0]):
+                out.append(Sequence(obs[i], act[i], rew[i], done[i], state[i],
+                                    float(prio[i]), obs.shape[1]))
         return out
 
     def publish_weights(self, actor_sd: Dict[str, torch.Tensor],

@@ -58,6 +58,23 @@ class StagingRing:
         return s
 
     @torch.no_grad()
+    def ingest_arrays(self, obs, act, rew, done, state, prio) -> int:
+        """Stage stacked numpy arrays (the fast path from poll_arrays)."""
+        total = 0
+        n_all = obs.shape[0]
+        for off in range(0, n_all, self.slot_seqs):
+            n = min(self.slot_seqs, n_all - off)
+            s = self._next_slot()
+            self._obs[s][:n].copy_(torch.from_numpy(obs[off:off + n]).to(self._obs[s].dtype))
+            self._act[s][:n].copy_(torch.from_numpy(act[off:off + n]))
+            self._rew[s][:n].copy_(torch.from_numpy(rew[off:off + n]))
+            self._done[s][:n].copy_(torch.from_numpy(done[off:off + n]))
+            self._state[s][:n].copy_(torch.from_numpy(state[off:off + n]))
+            self._prio[s][:n].copy_(torch.from_numpy(prio[off:off + n]))
+            total += self._dispatch(s, n)
+        return total
+
+    @torch.no_grad()
     def ingest(self, seqs: List[Sequence]) -> int:
         """Stage and enqueue sequences into the replay shard. Returns count."""
         total = 0
@@ -73,19 +90,23 @@ class StagingRing:
             self._done[s][:n].copy_(torch.from_numpy(np.stack([q.done for q in chunk])))
             self._state[s][:n].copy_(torch.from_numpy(np.stack([q.state for q in chunk])))
             self._prio[s][:n].copy_(torch.tensor([q.priority for q in chunk]))
-            if self.is_cuda:
-                # bulk H2D on the side stream, scatter on the learner stream
-                with torch.cuda.stream(self.stream):
-                    dev = [t[:n].to(self.device, non_blocking=True)
-                           for t in (self._obs[s], self._act[s], self._rew[s],
-                                     self._done[s], self._state[s], self._prio[s])]
-                    self._events[s].record(self.stream)
-                self._busy[s] = True
-                torch.cuda.current_stream().wait_event(self._events[s])
-                self.replay.add_batch(*dev)
-            else:
-                self.replay.add_batch(self._obs[s][:n], self._act[s][:n],
-                                      self._rew[s][:n], self._done[s][:n],
-                                      self._state[s][:n], self._prio[s][:n])
-            total += n
+            total += self._dispatch(s, n)
         return total
+
+    @torch.no_grad()
+    def _dispatch(self, s: int, n: int) -> int:
+        if self.is_cuda:
+            # bulk H2D on the side stream, scatter on the learner stream
+            with torch.cuda.stream(self.stream):
+                dev = [t[:n].to(self.device, non_blocking=True)
+                       for t in (self._obs[s], self._act[s], self._rew[s],
+                                 self._done[s], self._state[s], self._prio[s])]
+                self._events[s].record(self.stream)
+            self._busy[s] = True
+            torch.cuda.current_stream().wait_event(self._events[s])
+            self.replay.add_batch(*dev)
+        else:
+            self.replay.add_batch(self._obs[s][:n], self._act[s][:n],
+                                  self._rew[s][:n], self._done[s][:n],
+                                  self._state[s][:n], self._prio[s][:n])
+        return n

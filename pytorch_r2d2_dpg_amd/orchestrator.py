@@ -28,6 +28,17 @@ from .replay.staging import StagingRing
 from .utils.metrics import MetricsLogger
 
 
+def _px_err() -> int:
+    try:
+        import torch as _t
+        if _t.cuda.is_available():
+            from .ops import hip_ext
+            return int(hip_ext().lstm_px_error())
+    except Exception:  # noqa: BLE001
+        pass
+    return 0
+
+
 def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str] = None,
                  log_path: Optional[str] = None, max_steps: Optional[int] = None):
     dist = DistContext.from_env()
@@ -68,6 +79,11 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
     # (same stream) never sees a torn slot (§3.4 single-writer contract:
     # this thread is the only writer).
     stop_ingest = threading.Event()
+    sync_ingest = bool(os.environ.get("R2D2_SYNC_INGEST"))
+
+    def ingest_once():
+        for b in pool.poll_arrays():
+            staging.ingest_arrays(*b)
 
     def ingest_loop():
         if replay.is_cuda:
@@ -79,13 +95,17 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
             if not batches:
                 time.sleep(0.002)
 
-    ingest_thread = threading.Thread(target=ingest_loop, daemon=True)
-    ingest_thread.start()
+    ingest_thread = None
+    if not sync_ingest:
+        ingest_thread = threading.Thread(target=ingest_loop, daemon=True)
+        ingest_thread.start()
 
     steps = max_steps if max_steps is not None else cfg.learner.steps
     try:
         # ---- warm the replay
         while len(replay) < max(R.min_fill, cfg.learner.batch_size):
+            if sync_ingest:
+                ingest_once()
             pool.watchdog()
             time.sleep(0.05)
 
@@ -93,6 +113,8 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
         trans_acc = 0
         env_steps_last = pool.aggregate_env_steps()
         for step in range(learner.step_count, steps):
+            if sync_ingest:
+                ingest_once()
             batch = replay.sample(cfg.learner.batch_size)
             stats = learner.train_step(batch)
             trans_acc += stats.transitions
@@ -119,11 +141,13 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
                     eval_return=float(np.mean([r for _, r in pool.eval_returns[-10:]]))
                     if pool.eval_returns else 0.0,
                     actor_respawns=pool.respawn_count,
+                    px_err=_px_err(),
                 )
                 t_last, trans_acc, env_steps_last = now, 0, env_steps
     finally:
         stop_ingest.set()
-        ingest_thread.join(timeout=5)
+        if ingest_thread is not None:
+            ingest_thread.join(timeout=5)
         pool.stop()
         logger.close()
     return learner

@@ -103,7 +103,15 @@ class StagingRing:
                                  self._done[s], self._state[s], self._prio[s])]
                 self._events[s].record(self.stream)
             self._busy[s] = True
-            torch.cuda.current_stream().wait_event(self._events[s])
+            cur = torch.cuda.current_stream()
+            cur.wait_event(self._events[s])
+            # dev tensors are ALLOCATED on the side stream but CONSUMED by the
+            # scatter on this stream: without record_stream the allocator may
+            # recycle their memory for the next slot's H2D before the pending
+            # scatter reads it (observed as NaN training under the threaded
+            # ingest — silent replay corruption)
+            for t in dev:
+                t.record_stream(cur)
             self.replay.add_batch(*dev)
         else:
             self.replay.add_batch(self._obs[s][:n], self._act[s][:n],

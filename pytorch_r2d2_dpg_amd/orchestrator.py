@@ -82,17 +82,16 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
     sync_ingest = bool(os.environ.get("R2D2_SYNC_INGEST"))
 
     def ingest_once():
-        for b in pool.poll_arrays():
-            staging.ingest_arrays(*b)
+        pool.drain_episodes()
+        pool.ingest_rings(staging)
 
     def ingest_loop():
         if replay.is_cuda:
             torch.cuda.set_device(replay.device)
         while not stop_ingest.is_set():
-            batches = pool.poll_arrays()
-            for b in batches:
-                staging.ingest_arrays(*b)
-            if not batches:
+            pool.drain_episodes()
+            n = pool.ingest_rings(staging)
+            if not n:
                 time.sleep(0.002)
 
     ingest_thread = None

@@ -24,6 +24,7 @@ import torch.multiprocessing as mp
 from ..actor import ActorWorker, VectorActorWorker
 from ..config import Config
 from ..replay.sequence import Sequence
+from .seq_ring import SeqRing
 
 
 # ---------------------------------------------------------- sd <-> flat buf
@@ -54,7 +55,7 @@ def unpack_sd(template: Dict[str, torch.Tensor], buf: torch.Tensor) -> Dict[str,
 # -------------------------------------------------------------- actor proc
 def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver,
                 stop_event, heartbeat, crash_at: int, chunk: int = 64,
-                eval_mode: bool = False):
+                eval_mode: bool = False, ring=None):
     torch.set_num_threads(1)
     cfg = Config.from_dict(cfg_dict)
     E = max(cfg.actor.envs_per_actor, 1)
@@ -78,21 +79,10 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
             raise RuntimeError(f"fault injection: actor {actor_id} crash at {crash_at}")
         if eval_mode:
             seqs = []  # noise-free eval actors report episode returns only
-        if seqs:
-            # stacked-array payload: ONE pickle buffer per field instead of
-            # per-sequence tuples (unpickle cost on the learner side was the
-            # ingest bottleneck at high actor FPS)
-            payload = ("seqs",
-                       np.stack([s.obs for s in seqs]),
-                       np.stack([s.action for s in seqs]),
-                       np.stack([s.reward for s in seqs]),
-                       np.stack([s.done for s in seqs]),
-                       np.stack([s.state for s in seqs]),
-                       np.asarray([s.priority for s in seqs], dtype=np.float32))
-            try:
-                seq_queue.put(payload, timeout=5.0)
-            except queue_mod.Full:
-                pass  # backpressure: drop if learner can't keep up
+        for sq in seqs:
+            # zero-pickle data plane: write straight into the shared-memory
+            # ring (drop-newest on backpressure, as the old queue did)
+            ring.push(sq)
         new_eps = worker.episode_returns[ep_reported:]
         if new_eps:
             try:
@@ -127,6 +117,16 @@ class ActorPool:
         self._nc = flat_size(self._c_tpl)
         self.weight_buf = torch.zeros(self._na + self._nc, dtype=torch.float32)
         self.weight_buf.share_memory_()
+        # per-actor shared-memory sequence rings (zero-pickle data plane)
+        spec = probe.spec
+        L = cfg.replay.seq_len + cfg.replay.n_step
+        slot_bytes = (int(np.prod(spec.obs_shape)) * (1 if spec.pixel else 4)
+                      + spec.action_dim * 4 + 8) * L + cfg.model.hidden_size * 16
+        ring_slots = int(min(256, max(16, (32 << 20) // max(slot_bytes, 1))))
+        self.rings = [SeqRing(self._ctx, ring_slots, L, spec.obs_shape,
+                              np.uint8 if spec.pixel else np.float32,
+                              spec.action_dim, cfg.model.hidden_size)
+                      for _ in range(self.num_actors)]
         del probe
         self.procs: List[Optional[mp.Process]] = [None] * self.num_actors
         self._last_beat = [0] * self.num_actors
@@ -147,7 +147,7 @@ class ActorPool:
             args=(self.base_actor_id + i, self.cfg.to_dict(), self.seq_queue,
                   self.weight_buf, self.weight_ver, self.stop_event, self.heartbeat,
                   self.crash_at.get(self.base_actor_id + i, 0), self.chunk,
-                  i >= self._eval_from),
+                  i >= self._eval_from, self.rings[i]),
             daemon=True,
         )
         p.start()
@@ -165,25 +165,35 @@ class ActorPool:
                     p.terminate()
 
     # ------------------------------------------------------------- data i/o
-    def poll_arrays(self, max_batches: int = 64) -> List[tuple]:
-        """Drain stacked sequence batches (obs, act, rew, done, state, prio)
-        as numpy arrays — the fast ingest path (no per-sequence objects)."""
-        out: List[tuple] = []
-        for _ in range(max_batches):
+    def drain_episodes(self, max_msgs: int = 256) -> None:
+        """Drain the (light) episode-return queue."""
+        for _ in range(max_msgs):
             try:
                 payload = self.seq_queue.get_nowait()
             except queue_mod.Empty:
                 break
-            tag = payload[0]
-            if tag == "episodes":
-                _, aid, eps = payload
-                self.episode_returns.extend((aid, r) for r in eps)
-            elif tag == "eval_episodes":
-                _, aid, eps = payload
+            tag, aid, eps = payload
+            if tag == "eval_episodes":
                 self.eval_returns.extend((aid, r) for r in eps)
             else:
-                out.append(payload[1:])
-                self.total_sequences += payload[1].shape[0]
+                self.episode_returns.extend((aid, r) for r in eps)
+
+    def poll_arrays(self, max_batches: int = 64) -> List[tuple]:
+        """Drain the shared-memory rings as stacked numpy arrays (copies)."""
+        self.drain_episodes()
+        out: List[tuple] = []
+        for ring in self.rings:
+            ranges = ring.drain_ranges()
+            for lo, hi in ranges:
+                out.append((ring.obs[lo:hi].numpy().copy(),
+                            ring.act[lo:hi].numpy().copy(),
+                            ring.rew[lo:hi].numpy().copy(),
+                            ring.done[lo:hi].numpy().copy(),
+                            ring.state[lo:hi].numpy().copy(),
+                            ring.prio[lo:hi].numpy().copy()))
+                self.total_sequences += hi - lo
+            if ranges:
+                ring.commit()
         return out
 
     def poll(self, max_batches: int = 64) -> List[Sequence]:
@@ -195,6 +205,14 @@ class ActorPool:
                 out.append(Sequence(obs[i], act[i], rew[i], done[i], state[i],
                                     float(prio[i]), obs.shape[1]))
         return out
+
+    def ingest_rings(self, staging) -> int:
+        """Zero-pickle fast path: memcpy unread ring slots straight into the
+        pinned staging buffers. Called from the learner-side ingest thread."""
+        total = 0
+        for ring in self.rings:
+            total += staging.ingest_from_ring(ring)
+        return total
 
     def publish_weights(self, actor_sd: Dict[str, torch.Tensor],
                         critic_sd: Dict[str, torch.Tensor]) -> None:

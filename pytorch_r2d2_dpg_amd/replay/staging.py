@@ -58,6 +58,28 @@ class StagingRing:
         return s
 
     @torch.no_grad()
+    def ingest_from_ring(self, ring) -> int:
+        """Copy unread slots from a shared-memory SeqRing into pinned staging
+        (plain memcpy with dtype cast; no pickling)."""
+        ranges = ring.drain_ranges()
+        if not ranges:
+            return 0
+        total = 0
+        for lo, hi in ranges:
+            for off in range(lo, hi, self.slot_seqs):
+                n = min(self.slot_seqs, hi - off)
+                s = self._next_slot()
+                self._obs[s][:n].copy_(ring.obs[off:off + n])
+                self._act[s][:n].copy_(ring.act[off:off + n])
+                self._rew[s][:n].copy_(ring.rew[off:off + n])
+                self._done[s][:n].copy_(ring.done[off:off + n])
+                self._state[s][:n].copy_(ring.state[off:off + n])
+                self._prio[s][:n].copy_(ring.prio[off:off + n])
+                total += self._dispatch(s, n)
+        ring.commit()
+        return total
+
+    @torch.no_grad()
     def ingest_arrays(self, obs, act, rew, done, state, prio) -> int:
         """Stage stacked numpy arrays (the fast path from poll_arrays)."""
         total = 0

@@ -90,9 +90,8 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
             torch.cuda.set_device(replay.device)
         while not stop_ingest.is_set():
             pool.drain_episodes()
-            n = pool.ingest_rings(staging)
-            if not n:
-                time.sleep(0.002)
+            pool.ingest_rings(staging)
+            time.sleep(0.02)  # pace: batch ~dozens of sequences per pass
 
     ingest_thread = None
     if not sync_ingest:

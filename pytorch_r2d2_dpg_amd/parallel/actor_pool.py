@@ -209,10 +209,7 @@ class ActorPool:
     def ingest_rings(self, staging) -> int:
         """Zero-pickle fast path: memcpy unread ring slots straight into the
         pinned staging buffers. Called from the learner-side ingest thread."""
-        total = 0
-        for ring in self.rings:
-            total += staging.ingest_from_ring(ring)
-        return total
+        return staging.ingest_from_rings(self.rings)
 
     def publish_weights(self, actor_sd: Dict[str, torch.Tensor],
                         critic_sd: Dict[str, torch.Tensor]) -> None:

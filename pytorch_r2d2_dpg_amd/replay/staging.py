@@ -58,25 +58,39 @@ class StagingRing:
         return s
 
     @torch.no_grad()
-    def ingest_from_ring(self, ring) -> int:
-        """Copy unread slots from a shared-memory SeqRing into pinned staging
-        (plain memcpy with dtype cast; no pickling)."""
-        ranges = ring.drain_ranges()
-        if not ranges:
-            return 0
+    def ingest_from_rings(self, rings) -> int:
+        """Copy unread slots from ALL shared-memory SeqRings into pinned
+        staging, AGGREGATED into full slots before each dispatch — one
+        dispatch is ~15 CUDA enqueues, and thousands of tiny dispatches/s
+        from the ingest thread contend with the learner on the HIP runtime
+        lock (plain memcpy with dtype cast; no pickling)."""
         total = 0
-        for lo, hi in ranges:
-            for off in range(lo, hi, self.slot_seqs):
-                n = min(self.slot_seqs, hi - off)
-                s = self._next_slot()
-                self._obs[s][:n].copy_(ring.obs[off:off + n])
-                self._act[s][:n].copy_(ring.act[off:off + n])
-                self._rew[s][:n].copy_(ring.rew[off:off + n])
-                self._done[s][:n].copy_(ring.done[off:off + n])
-                self._state[s][:n].copy_(ring.state[off:off + n])
-                self._prio[s][:n].copy_(ring.prio[off:off + n])
-                total += self._dispatch(s, n)
-        ring.commit()
+        s = None
+        fill = 0
+        for ring in rings:
+            ranges = ring.drain_ranges()
+            for lo, hi in ranges:
+                off = lo
+                while off < hi:
+                    if s is None:
+                        s = self._next_slot()
+                        fill = 0
+                    n = min(hi - off, self.slot_seqs - fill)
+                    self._obs[s][fill:fill + n].copy_(ring.obs[off:off + n])
+                    self._act[s][fill:fill + n].copy_(ring.act[off:off + n])
+                    self._rew[s][fill:fill + n].copy_(ring.rew[off:off + n])
+                    self._done[s][fill:fill + n].copy_(ring.done[off:off + n])
+                    self._state[s][fill:fill + n].copy_(ring.state[off:off + n])
+                    self._prio[s][fill:fill + n].copy_(ring.prio[off:off + n])
+                    fill += n
+                    off += n
+                    if fill == self.slot_seqs:
+                        total += self._dispatch(s, fill)
+                        s = None
+            if ranges:
+                ring.commit()
+        if s is not None and fill:
+            total += self._dispatch(s, fill)
         return total
 
     @torch.no_grad()

@@ -1,9 +1,10 @@
 """Actor process pool (SURVEY.md §3.1/§5 failure detection).
 
-Spawns N CPU actor processes per learner rank. Each actor runs the rollout
-loop (actor.py), pushes cut sequences through a queue, and pulls fresh
-weights from a shared-memory flat parameter buffer (version-stamped, no
-per-sync pickling of state_dicts).
+Spawns N CPU actor processes per learner rank (each optionally vectorized
+over K envs). Each actor runs the rollout loop (actor.py), writes cut
+sequences into its zero-pickle shared-memory ring (seq_ring.py; episode
+returns go through a light queue), and pulls fresh weights from a
+shared-memory flat parameter buffer (version-stamped).
 
 Elasticity: actors are stateless beyond their env, so the pool heartbeats
 each actor (env-step watermark) and respawns dead/stalled ones with a fresh

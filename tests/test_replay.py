@@ -172,3 +172,43 @@ def test_replay_ring_wraps():
     # slots 0,1 overwritten by 4,5
     assert rep.obs[0, 0, 0].item() == 4.0
     assert rep.obs[1, 0, 0].item() == 5.0
+
+
+# ------------------------------------------------------------- shm seq ring
+def test_seq_ring_push_drain_commit():
+    import multiprocessing as mp
+    from pytorch_r2d2_dpg_amd.parallel.seq_ring import SeqRing
+
+    ctx = mp.get_context("spawn")
+    ring = SeqRing(ctx, slots=4, stored_len=6, obs_shape=(3,), obs_dtype=np.float32,
+                   action_dim=2, hidden_size=8)
+    for i in range(3):
+        assert ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8, prio=1.0 + i, fill=float(i)))
+    ranges = ring.drain_ranges()
+    assert ranges == [(0, 3)]
+    assert ring.obs[1, 0, 0].item() == 1.0
+    ring.commit()
+    # wrap-around: push 4 more -> ranges split in two
+    for i in range(4):
+        assert ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8, fill=10.0 + i))
+    ranges = ring.drain_ranges()
+    assert ranges == [(3, 4), (0, 3)]
+    assert ring.obs[3, 0, 0].item() == 10.0
+    assert ring.obs[0, 0, 0].item() == 11.0
+    ring.commit()
+
+
+def test_seq_ring_backpressure_drops_newest():
+    import multiprocessing as mp
+    from pytorch_r2d2_dpg_amd.parallel.seq_ring import SeqRing
+
+    ctx = mp.get_context("spawn")
+    ring = SeqRing(ctx, slots=2, stored_len=6, obs_shape=(3,), obs_dtype=np.float32,
+                   action_dim=2, hidden_size=8)
+    assert ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8))
+    assert ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8))
+    assert not ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8))  # full -> dropped
+    assert ring.dropped.value == 1
+    ring.drain_ranges()
+    ring.commit()
+    assert ring.push(_mk_seq(L=6, obs_dim=3, A=2, H=8))  # space again

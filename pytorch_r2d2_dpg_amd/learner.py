@@ -33,7 +33,7 @@ import torch
 from .config import Config
 from .models import make_models
 from .ops.polyak import PolyakUpdater
-from .ops.lstm import lstm_seq_pair
+from .ops.lstm import lstm_seq_pair, lstm_seq_pair_halfgrad
 from .ops.td import td_targets_and_priorities
 from .parallel.dist import DistContext
 
@@ -215,14 +215,29 @@ class Learner:
 
         self._pe("burn_in", 1)
         self._pe("target", 0)
-        # ---- target unroll + critic online fwd (main stream; balances the
-        # side stream's actor-chain length)
+        # ---- target-actor unroll + critic online fwd in ONE paired
+        # column-split launch (the two are independent: the online critic
+        # reads the REPLAY action, not the target action), then the
+        # target-critic unroll which consumes mu_t (main stream; balances
+        # the side stream's actor-chain length)
+        ta = self.target_actor
         with torch.no_grad():
-            mu_t, _ = self.target_actor(obs[bi:bi + Tn], (h_a2, c_a2))
+            feat_ta = ta.trunk(obs[bi:bi + Tn])
+        feat_cf = self.critic.trunk(obs[bi:bi + T])
+        x_cf = torch.cat([feat_cf, action[bi:bi + T]], dim=-1)
+        hs_ta, _, _, hs_cf, _, _ = lstm_seq_pair_halfgrad(
+            feat_ta, h_a2, c_a2,
+            ta.lstm.weight_x, ta.lstm.weight_h, ta.lstm.bias,
+            x_cf, h_c2.detach(), c_c2.detach(),
+            self.critic.lstm.weight_x, self.critic.lstm.weight_h,
+            self.critic.lstm.bias)
+        Hs = ta.hidden_size
+        with torch.no_grad():
+            mu_t = (torch.tanh(ta.head(hs_ta.reshape(-1, Hs)))
+                    .view(Tn, -1, ta.action_dim) * ta.action_scale)
             qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
             q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
-        q_online, _ = self.critic(obs[bi:bi + T], action[bi:bi + T],
-                                  (h_c2.detach(), c_c2.detach()))
+        q_online = self.critic.head(hs_cf.reshape(-1, Hs)).view(T, -1)
         q_online_bt = q_online.transpose(0, 1).float()  # [B, T]
         with torch.no_grad():
             y, td_abs, prio = td_targets_and_priorities(

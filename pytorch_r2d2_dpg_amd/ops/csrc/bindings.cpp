@@ -40,10 +40,12 @@ extern "C" int launch_lstm_fwd_px_bf16(const void*, const void*, const void*, co
                                        void*, void*, void*, float*, unsigned long long*,
                                        unsigned int*, int, int, int, int, hipStream_t);
 extern "C" int launch_lstm_fwd_px2_bf16(const void*, const void*, const void*, const void*,
-                                        void*, void*, unsigned long long*,
+                                        void*, void*, void*, float*, int, int,
+                                        unsigned long long*,
                                         const void*, const void*, const void*, const void*,
-                                        void*, void*, unsigned long long*,
-                                        unsigned int*, int, int, int, hipStream_t);
+                                        void*, void*, void*, float*, int, int,
+                                        unsigned long long*,
+                                        unsigned int*, int, int, hipStream_t);
 extern "C" int launch_lstm_bwd_px_bf16(const void*, const void*, const void*, const void*,
                                        const float*, const void*, const void*, void*, void*,
                                        void*, unsigned char*, unsigned long long*,
@@ -185,37 +187,52 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
     return {h_seq, c_final, gates_act, c_seq};
 }
 
-// ---- K1-pair: two independent no-grad LSTM unrolls in ONE launch ----------
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+// ---- K1-pair: two independent LSTM unrolls in ONE launch ------------------
+// Same B/H; T and stash may differ per set (set with stash=true gets its
+// activations + fp32 cell states written for the K2 backward — used to pair
+// the no-grad target-actor unroll with the grad-tracked online critic fwd).
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+           torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
                   torch::Tensor wh0, torch::Tensor gx1, torch::Tensor h01,
-                  torch::Tensor c01, torch::Tensor wh1) {
+                  torch::Tensor c01, torch::Tensor wh1,
+                  bool stash0 = false, bool stash1 = false) {
     for (auto* t : {&gx0, &h00, &c00, &wh0, &gx1, &h01, &c01, &wh1})
         check_dev_contig(*t, "pair arg");
-    const auto T = gx0.size(0);
+    const auto T0 = gx0.size(0);
+    const auto T1 = gx1.size(0);
     const auto B = gx0.size(1);
     const auto H = h00.size(1);
-    TORCH_CHECK(gx1.size(0) == T && gx1.size(1) == B && h01.size(1) == H,
-                "pair shape mismatch");
+    TORCH_CHECK(gx1.size(1) == B && h01.size(1) == H, "pair shape mismatch");
     TORCH_CHECK(gx0.scalar_type() == torch::kBFloat16, "pair path is bf16-only");
     auto wp0 = pack_fragment_major(wh0);
     auto wp1 = pack_fragment_major(wh1);
     const int RT = (int)((B + 15) / 16);
     auto hx0 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
     auto hx1 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
-    auto hs0 = torch::empty({T, B, H}, gx0.options());
-    auto hs1 = torch::empty({T, B, H}, gx0.options());
+    auto hs0 = torch::empty({T0, B, H}, gx0.options());
+    auto hs1 = torch::empty({T1, B, H}, gx0.options());
     auto cf0 = torch::empty({B, H}, gx0.options());
     auto cf1 = torch::empty({B, H}, gx0.options());
+    auto ga0 = torch::empty({stash0 ? T0 : 0, B, 4 * H}, gx0.options());
+    auto ga1 = torch::empty({stash1 ? T1 : 0, B, 4 * H}, gx0.options());
+    auto cs0 = torch::empty({stash0 ? T0 : 0, B, H},
+                            gx0.options().dtype(torch::kFloat32));
+    auto cs1 = torch::empty({stash1 ? T1 : 0, B, H},
+                            gx0.options().dtype(torch::kFloat32));
     auto& err = px_err_word(gx0.device());
     int rc = launch_lstm_fwd_px2_bf16(
         gx0.data_ptr(), h00.data_ptr(), c00.data_ptr(), wp0.data_ptr(),
-        hs0.data_ptr(), cf0.data_ptr(), (unsigned long long*)hx0.data_ptr<int64_t>(),
+        hs0.data_ptr(), cf0.data_ptr(), ga0.data_ptr(),
+        stash0 ? cs0.data_ptr<float>() : nullptr, stash0 ? 1 : 0, (int)T0,
+        (unsigned long long*)hx0.data_ptr<int64_t>(),
         gx1.data_ptr(), h01.data_ptr(), c01.data_ptr(), wp1.data_ptr(),
-        hs1.data_ptr(), cf1.data_ptr(), (unsigned long long*)hx1.data_ptr<int64_t>(),
-        (unsigned int*)err.data_ptr<int32_t>(), (int)T, (int)B, (int)H, cur_stream());
+        hs1.data_ptr(), cf1.data_ptr(), ga1.data_ptr(),
+        stash1 ? cs1.data_ptr<float>() : nullptr, stash1 ? 1 : 0, (int)T1,
+        (unsigned long long*)hx1.data_ptr<int64_t>(),
+        (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, cur_stream());
     TORCH_CHECK(rc == 0, "pair geometry unsupported (fall back to two calls)");
-    return {hs0, cf0, hs1, cf1};
+    return {hs0, cf0, ga0, cs0, hs1, cf1, ga1, cs1};
 }
 
 // ---- K2: fused LSTM sequence backward -------------------------------------
@@ -390,7 +407,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_px_error", &lstm_px_error,
           "read-and-clear the column-split LSTM spin-timeout error word");
     m.def("lstm_seq_fwd_pair", &lstm_seq_fwd_pair,
-          "two independent no-grad LSTM unrolls in one launch");
+          "two independent LSTM unrolls in one launch (per-set T/stash)",
+          py::arg("gx0"), py::arg("h00"), py::arg("c00"), py::arg("wh0"),
+          py::arg("gx1"), py::arg("h01"), py::arg("c01"), py::arg("wh1"),
+          py::arg("stash0") = false, py::arg("stash1") = false);
     m.def("set_px_serialize", &set_px_serialize,
           "serialize column-split kernels device-wide (off by default)");
 }

@@ -514,11 +514,13 @@ extern "C" int launch_lstm_bwd_px_bf16(
 }
 
 // ------------------------------------------------- paired (z=2) forward
-// Two INDEPENDENT same-shape LSTMs (e.g. the actor and critic burn-in
-// unrolls) in one launch: grid.z selects the pointer set. A single grid
-// dispatches uniformly, so the two halves co-reside by construction —
-// unlike two concurrent px launches on different streams, which contend
-// for CUs and run ~3x slower than back-to-back.
+// Two INDEPENDENT LSTM unrolls (same B and H; T and stash may differ per
+// set — e.g. the actor/critic burn-ins, or the target-actor unroll paired
+// with the grad-tracked online critic forward) in one launch: grid.z
+// selects the pointer set. A single grid dispatches uniformly, so the two
+// halves co-reside by construction — unlike two concurrent px launches on
+// different streams, which contend for CUs and run ~3x slower than
+// back-to-back.
 struct PxArgs {
     const __hip_bfloat16* gates_x;
     const __hip_bfloat16* h0;
@@ -526,13 +528,18 @@ struct PxArgs {
     const __hip_bfloat16* wh;
     __hip_bfloat16* h_seq;
     __hip_bfloat16* c_final;
+    __hip_bfloat16* gates_act;  // stash only
+    float* c_seq;               // stash only
     unsigned long long* hx;
+    int T;
+    int stash;
 };
 
 template <int CG>
 __global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
-lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int T, int B, int H) {
+lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
     const PxArgs& a = blockIdx.z ? a1 : a0;
+    const int T = a.T;
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
     const int tid = threadIdx.x;
@@ -626,6 +633,8 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int T, int B, int H
         __syncthreads();
 
         __hip_bfloat16* hs_t = a.h_seq + (long)t * B * H;
+        __hip_bfloat16* ga_t = a.stash ? a.gates_act + (long)t * B * H4 : nullptr;
+        float* cs_t = a.stash ? a.c_seq + (long)t * B * H : nullptr;
         char* rd_buf = smem + (t & 1) * tile_bytes;
         for (int kk = 0; kk < H; kk += 32) {
             int k = kk + a_kgrp;
@@ -666,7 +675,16 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int T, int B, int H
                     __hip_atomic_store(&slot[row * (H / 2) + col / 2],
                                        ((unsigned long long)tag << 32) | lo, RLX_AGENT);
                 }
-                if (grow < B) hs_t[(long)grow * H + col] = f2bf(hn);
+                if (grow < B) {
+                    hs_t[(long)grow * H + col] = f2bf(hn);
+                    if (a.stash) {
+                        cs_t[(long)grow * H + col] = c;
+                        ga_t[(long)grow * H4 + 0 * H + col] = f2bf(gi);
+                        ga_t[(long)grow * H4 + 1 * H + col] = f2bf(gf);
+                        ga_t[(long)grow * H4 + 2 * H + col] = f2bf(gg);
+                        ga_t[(long)grow * H4 + 3 * H + col] = f2bf(go);
+                    }
+                }
             }
         }
         __syncthreads();
@@ -681,10 +699,12 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int T, int B, int H
 
 extern "C" int launch_lstm_fwd_px2_bf16(
     const void* gx0, const void* h00, const void* c00, const void* wh0,
-    void* hs0, void* cf0, unsigned long long* hx0,
+    void* hs0, void* cf0, void* ga0, float* cs0, int stash0, int T0,
+    unsigned long long* hx0,
     const void* gx1, const void* h01, const void* c01, const void* wh1,
-    void* hs1, void* cf1, unsigned long long* hx1,
-    unsigned int* err, int T, int B, int H, hipStream_t stream) {
+    void* hs1, void* cf1, void* ga1, float* cs1, int stash1, int T1,
+    unsigned long long* hx1,
+    unsigned int* err, int B, int H, hipStream_t stream) {
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
     int CG = 0;
@@ -702,14 +722,16 @@ extern "C" int launch_lstm_fwd_px2_bf16(
     size_t lds = (size_t)2 * 16 * H * 2;
     PxArgs a0{(const __hip_bfloat16*)gx0, (const __hip_bfloat16*)h00,
               (const __hip_bfloat16*)c00, (const __hip_bfloat16*)wh0,
-              (__hip_bfloat16*)hs0, (__hip_bfloat16*)cf0, hx0};
+              (__hip_bfloat16*)hs0, (__hip_bfloat16*)cf0,
+              (__hip_bfloat16*)ga0, cs0, hx0, T0, stash0};
     PxArgs a1{(const __hip_bfloat16*)gx1, (const __hip_bfloat16*)h01,
               (const __hip_bfloat16*)c01, (const __hip_bfloat16*)wh1,
-              (__hip_bfloat16*)hs1, (__hip_bfloat16*)cf1, hx1};
+              (__hip_bfloat16*)hs1, (__hip_bfloat16*)cf1,
+              (__hip_bfloat16*)ga1, cs1, hx1, T1, stash1};
     switch (CG) {
-        case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
-        case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
-        case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, T, B, H); break;
+        case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+        case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+        case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
         default: return -1;
     }
     return 0;

@@ -98,22 +98,87 @@ def lstm_seq(
 
 
 def lstm_seq_pair(x0, h00, c00, wx0, wh0, b0, x1, h01, c01, wx1, wh1, b1):
-    """Two independent no-grad LSTM unrolls (same T, B, H) in ONE launch
+    """Two independent no-grad LSTM unrolls (same B, H) in ONE launch
     (burn-in actor + critic). Falls back to two sequential calls when the
     paired geometry is unsupported. Returns (h_seq0, hT0, cT0, h_seq1, hT1, cT1)."""
     if x0.is_cuda and x0.dtype == torch.bfloat16 and not torch.is_grad_enabled():
-        T, B, _ = x0.shape
+        T0, B, _ = x0.shape
+        T1 = x1.shape[0]
         H = h00.shape[1]
         ext = hip_ext()
-        gx0 = torch.addmm(b0, x0.reshape(T * B, -1), wx0.t()).view(T, B, 4 * H)
-        gx1 = torch.addmm(b1, x1.reshape(T * B, -1), wx1.t()).view(T, B, 4 * H)
+        gx0 = torch.addmm(b0, x0.reshape(T0 * B, -1), wx0.t()).view(T0, B, 4 * H)
+        gx1 = torch.addmm(b1, x1.reshape(T1 * B, -1), wx1.t()).view(T1, B, 4 * H)
         try:
-            hs0, cf0, hs1, cf1 = ext.lstm_seq_fwd_pair(
+            hs0, cf0, _, _, hs1, cf1, _, _ = ext.lstm_seq_fwd_pair(
                 gx0.contiguous(), h00.contiguous(), c00.contiguous(), wh0.contiguous(),
                 gx1.contiguous(), h01.contiguous(), c01.contiguous(), wh1.contiguous())
             return hs0, hs0[-1], cf0, hs1, hs1[-1], cf1
         except RuntimeError:
             pass
     a = lstm_seq(x0, h00, c00, wx0, wh0, b0)
+    c = lstm_seq(x1, h01, c01, wx1, wh1, b1)
+    return a[0], a[1], a[2], c[0], c[1], c[2]
+
+
+class _FusedLSTMSeqPairHalf(torch.autograd.Function):
+    """Paired launch where set0 is a NO-GRAD unroll (precomputed gates_x) and
+    set1 is grad-tracked with stashed activations. Backward is identical to
+    _FusedLSTMSeq's for set1; set0's outputs are marked non-differentiable."""
+
+    @staticmethod
+    def forward(ctx, x1, h01, c01, wx1, wh1, b1, gx0, h00, c00, wh0):
+        T1, B, I1 = x1.shape
+        H = h01.shape[1]
+        ext = hip_ext()
+        gx1 = torch.addmm(b1, x1.reshape(T1 * B, I1), wx1.t()).view(T1, B, 4 * H)
+        hs0, cf0, _, _, hs1, cf1, ga1, cs1 = ext.lstm_seq_fwd_pair(
+            gx0.contiguous(), h00.contiguous(), c00.contiguous(), wh0.contiguous(),
+            gx1.contiguous(), h01.contiguous(), c01.contiguous(), wh1.contiguous(),
+            False, True)
+        ctx.save_for_backward(x1, h01, c01, wx1, wh1, hs1, ga1, cs1)
+        ctx.mark_non_differentiable(hs0, cf0)
+        return hs0, cf0, hs1, cf1
+
+    @staticmethod
+    def backward(ctx, dhs0, dcf0, dh_seq, dcT):
+        x, h0, c0, wx, wh, h_seq, gates_act, c_seq = ctx.saved_tensors
+        T, B, I = x.shape
+        H = h0.shape[1]
+        ext = hip_ext()
+        zero_hT = torch.zeros_like(h0)
+        whT = wh.t().contiguous()
+        dgates, dh0, dc0 = ext.lstm_seq_bwd(
+            dh_seq.contiguous(), zero_hT, dcT.contiguous(), gates_act, c_seq,
+            c0.contiguous(), whT,
+        )
+        dg_flat = dgates.reshape(T * B, 4 * H)
+        dx = (dg_flat @ wx).view(T, B, I)
+        dwx = dg_flat.t() @ x.reshape(T * B, I)
+        h_prev = torch.cat([h0.unsqueeze(0), h_seq[:-1]], dim=0).reshape(T * B, H)
+        dwh = dg_flat.t() @ h_prev
+        db = dg_flat.float().sum(dim=0).to(dg_flat.dtype)
+        return dx, dh0, dc0, dwx, dwh, db, None, None, None, None
+
+
+def lstm_seq_pair_halfgrad(x0, h00, c00, wx0, wh0, b0, x1, h01, c01, wx1, wh1, b1):
+    """Set0 runs WITHOUT grad (e.g. the target-actor unroll), set1 WITH grad
+    (the online critic forward), in one paired column-split launch; T0 and T1
+    may differ. Falls back to two sequential ``lstm_seq`` calls on CPU or
+    unsupported geometry. Returns (h_seq0, hT0, cT0, h_seq1, hT1, cT1)."""
+    if (x0.is_cuda and x0.dtype == torch.bfloat16 and torch.is_grad_enabled()
+            and x1.requires_grad):
+        T0, B, I0 = x0.shape
+        H = h00.shape[1]
+        with torch.no_grad():
+            gx0 = torch.addmm(b0, x0.reshape(T0 * B, I0), wx0.t()).view(T0, B, 4 * H)
+        try:
+            hs0, cf0, hs1, cf1 = _FusedLSTMSeqPairHalf.apply(
+                x1, h01, c01, wx1, wh1, b1,
+                gx0, h00.detach(), c00.detach(), wh0)
+            return hs0, hs0[-1], cf0, hs1, hs1[-1], cf1
+        except RuntimeError:
+            pass
+    with torch.no_grad():
+        a = lstm_seq(x0, h00.detach(), c00.detach(), wx0, wh0, b0)
     c = lstm_seq(x1, h01, c01, wx1, wh1, b1)
     return a[0], a[1], a[2], c[0], c[1], c[2]

@@ -376,3 +376,45 @@ def test_lstm_fwd_pair_matches_singles():
     assert ext().lstm_px_error() == 0
     assert torch.equal(hs0, r0[0]) and torch.equal(cT0, r0[2])
     assert torch.equal(hs1, r1[0]) and torch.equal(cT1, r1[2])
+
+
+def test_lstm_pair_halfgrad_matches_singles():
+    """Half-grad pair (no-grad target unroll + grad-tracked critic fwd, T
+    differs per set) must match the unpaired path bitwise in forward and
+    gradients (same kernels, same order of GEMMs in backward)."""
+    B, I, H = 64, 40, 512
+    T0, T1 = 13, 9  # per-set T
+    g = torch.Generator().manual_seed(33)
+
+    def mk(T):
+        x = torch.randn(T, B, I, generator=g).cuda().bfloat16()
+        h0 = (torch.randn(B, H, generator=g) * 0.3).cuda().bfloat16()
+        c0 = (torch.randn(B, H, generator=g) * 0.3).cuda().bfloat16()
+        wx = (torch.randn(4 * H, I, generator=g) / I ** 0.5).cuda().bfloat16()
+        wh = (torch.randn(4 * H, H, generator=g) / H ** 0.5).cuda().bfloat16()
+        b = (torch.randn(4 * H, generator=g) * 0.1).cuda().bfloat16()
+        return x, h0, c0, wx, wh, b
+
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq, lstm_seq_pair_halfgrad
+    a = mk(T0)
+    c_base = mk(T1)
+
+    def grad_set():
+        return tuple(t.clone().requires_grad_(i not in (1, 2))
+                     for i, t in enumerate(c_base))
+
+    c1, c2 = grad_set(), grad_set()
+    hs0, hT0, cT0, hs1, hT1, cT1 = lstm_seq_pair_halfgrad(*a, *c1)
+    (hs1.float().square().mean() + hT1.float().abs().mean()).backward()
+    with torch.no_grad():
+        r0 = lstm_seq(*a)
+    h_r, hT_r, cT_r = lstm_seq(*c2)
+    (h_r.float().square().mean() + hT_r.float().abs().mean()).backward()
+    torch.cuda.synchronize()
+    assert ext().lstm_px_error() == 0
+    assert torch.equal(hs0, r0[0]) and torch.equal(cT0, r0[2])
+    assert torch.equal(hs1, h_r) and torch.equal(cT1, cT_r)
+    for name, p1, p2 in zip(("x", "wx", "wh", "b"),
+                            (c1[0], c1[3], c1[4], c1[5]),
+                            (c2[0], c2[3], c2[4], c2[5])):
+        assert p1.grad is not None and torch.equal(p1.grad, p2.grad), name

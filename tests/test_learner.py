@@ -115,3 +115,57 @@ def test_config_toml_roundtrip(tmp_path):
     cfg.apply_overrides(["learner.batch_size=32", "model.dtype=bf16"])
     assert cfg.learner.batch_size == 32
     assert cfg.model.dtype == "bf16"
+
+
+def test_manual_decomposition_matches_module_forward():
+    """learner._step_core computes the target-actor and online-critic
+    forwards manually (trunk -> paired LSTM -> head) so the two unrolls can
+    share one px2 launch; this invariant keeps that decomposition in
+    lock-step with ActorNet.forward / CriticNet.forward."""
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.models import make_models
+    from pytorch_r2d2_dpg_amd.ops.lstm import lstm_seq, lstm_seq_pair_halfgrad
+
+    spec = make_env("pendulum").spec
+    torch.manual_seed(0)
+
+    class _M:
+        hidden_size, trunk_sizes, conv_channels = 64, (64,), (32, 64, 64)
+    actor, critic, ta, tc = make_models(spec, _M)
+    T, B, H = 7, 5, 64
+    obs = torch.randn(T, B, *spec.obs_shape)
+    act = torch.randn(T, B, spec.action_dim).clamp_(-1, 1)
+    h = torch.randn(B, H) * 0.1
+    c = torch.randn(B, H) * 0.1
+
+    # actor: module forward vs manual head math
+    mu_mod, _ = ta(obs, (h, c))
+    feat = ta.trunk(obs)
+    hs, _, _ = lstm_seq(feat, h, c, ta.lstm.weight_x, ta.lstm.weight_h, ta.lstm.bias)
+    mu_man = (torch.tanh(ta.head(hs.reshape(-1, H))).view(T, B, -1)
+              * ta.action_scale)
+    assert torch.equal(mu_mod, mu_man)
+
+    # critic: module forward vs manual (cat -> lstm -> head)
+    q_mod, _ = critic(obs, act, (h, c))
+    x = torch.cat([critic.trunk(obs), act], dim=-1)
+    hs_c, _, _ = lstm_seq(x, h, c, critic.lstm.weight_x, critic.lstm.weight_h,
+                          critic.lstm.bias)
+    q_man = critic.head(hs_c.reshape(-1, H)).view(T, B)
+    assert torch.equal(q_mod, q_man)
+
+    # CPU fallback of the half-grad pair == two plain lstm_seq calls,
+    # gradients included
+    feat2 = feat.detach()
+    x1 = x.detach().requires_grad_(True)
+    hs0, hT0, cT0, hs1, hT1, cT1 = lstm_seq_pair_halfgrad(
+        feat2, h, c, ta.lstm.weight_x, ta.lstm.weight_h, ta.lstm.bias,
+        x1, h, c, critic.lstm.weight_x, critic.lstm.weight_h, critic.lstm.bias)
+    assert torch.equal(hs0, hs)
+    hs1.square().mean().backward()
+    x2 = x.detach().requires_grad_(True)
+    r, _, _ = lstm_seq(x2, h, c, critic.lstm.weight_x, critic.lstm.weight_h,
+                       critic.lstm.bias)
+    assert torch.equal(hs1, r)
+    r.square().mean().backward()
+    assert torch.allclose(x1.grad, x2.grad)

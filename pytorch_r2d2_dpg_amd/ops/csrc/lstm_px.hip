@@ -340,7 +340,10 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
         const float* cs_p = (t > 0) ? c_seq + (long)(t - 1) * B * H : nullptr;
         __hip_bfloat16* dg_t = dgates + (long)t * B * H4;
 
-        // ---- lane-local dgates for own columns -> LDS own region + global
+        // ---- lane-local dgates for own columns -> LDS own region (the
+        // global dgates write is DEFERRED until after the flag publish:
+        // the sc1 drain below is vmcnt(0), so any outstanding global
+        // stores here would delay every peer's flag wait each timestep)
         {
             const int col = col_base + c_col;
 #pragma unroll
@@ -365,10 +368,6 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
                     dgg = dc * gi * (1.0f - gg * gg);
                     dgo = do_ * go * (1.0f - go);
                     dc_reg[r] = dc * gf;
-                    dg_t[(long)grow * H4 + 0 * H + col] = f2bf(dgi);
-                    dg_t[(long)grow * H4 + 1 * H + col] = f2bf(dgf);
-                    dg_t[(long)grow * H4 + 2 * H + col] = f2bf(dgg);
-                    dg_t[(long)grow * H4 + 3 * H + col] = f2bf(dgo);
                 } else {
                     dc_reg[r] = 0.0f;
                 }
@@ -406,6 +405,26 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
             if (tid == 0)
                 __hip_atomic_store(&flg[((t & 1) * gridDim.x + rt) * CG + cg],
                                    ((unsigned long long)tag << 32) | 1u, RLX_AGENT);
+        }
+
+        // ---- deferred global dgates write (own columns, read back from the
+        // LDS tile — bitwise the same bf16 values): fire-and-forget stores
+        // that overlap the peer-flag wait + slab sweep and have the whole
+        // next timestep to retire before the next vmcnt(0) drain
+        {
+            const int col = col_base + c_col;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = c_row0 + r;
+                int grow = brow0 + row;
+                if (grow < B) {
+#pragma unroll
+                    for (int g = 0; g < 4; ++g)
+                        dg_t[(long)grow * H4 + g * H + col] =
+                            *reinterpret_cast<const __hip_bfloat16*>(
+                                &buf[px_swz(row * rowbytes4 + (g * H + col) * 2, row)]);
+                }
+            }
         }
 
         // ---- wait peers' flags, then sweep their slab regions into LDS

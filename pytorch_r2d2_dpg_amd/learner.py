@@ -113,9 +113,15 @@ class Learner:
         # chains, +20% measured); at larger batches two concurrent
         # column-split grids contend for CU residency and crawl, so run the
         # chains sequentially there (the px kernels fill the chip anyway).
-        # R2D2_NO_OVERLAP=1 forces off.
+        # Also gated on train_len <= 48: at B=256 the two concurrent bwd-px
+        # grids are an exact 256-CU fit, and with the longer kernels of
+        # seq-120 configs the collision window with stray kernels (ingest
+        # scatter, GEMMs) is wide enough that displaced blocks spin out
+        # (observed as multi-second steps); sequential chains are safe and
+        # near-parity there. R2D2_NO_OVERLAP=1 forces off.
         overlap_ok = (self.device.type == "cuda"
                       and cfg.learner.batch_size <= 256
+                      and self.train_len <= 48
                       and not _os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()

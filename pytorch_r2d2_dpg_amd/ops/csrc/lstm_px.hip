@@ -122,8 +122,8 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 200000000u) {
-                    // bounded spin: flag the error and PROCEED with stale
+                if (++spins > 20000000u) {
+                    // bounded spin (~0.5 s at s_sleep granularity): flag the error and PROCEED with stale
                     // data — control flow stays convergent so no thread
                     // ever skips a barrier (a diverged exit deadlocks the
                     // block and hangs the device)
@@ -434,7 +434,7 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
                 unsigned long long v = __hip_atomic_load(f, RLX_AGENT);
                 unsigned spins = 0;
                 while ((unsigned)(v >> 32) != tag) {
-                    if (++spins > 200000000u) {  // convergent bail (see fwd)
+                    if (++spins > 20000000u) {  // convergent bail (see fwd)
                         __hip_atomic_store((gu32*)err, 0x10000u + (unsigned)t, RLX_AGENT);
                         break;
                     }
@@ -625,7 +625,7 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
                 pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 200000000u) {  // convergent bail (see single fwd)
+                if (++spins > 20000000u) {  // convergent bail (see single fwd)
                     __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
                     break;
                 }

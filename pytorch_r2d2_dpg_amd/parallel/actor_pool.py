@@ -71,10 +71,12 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
     ep_reported = 0
     while not stop_event.is_set():
         ver = int(weight_ver.value)
-        if ver != last_ver and ver > 0:
-            worker.load_weights(unpack_sd(a_tpl, weight_buf[:na]),
-                                unpack_sd(c_tpl, weight_buf[na:]))
-            last_ver = ver
+        if ver != last_ver and ver > 0 and ver % 2 == 0:
+            a_sd = unpack_sd(a_tpl, weight_buf[:na])
+            c_sd = unpack_sd(c_tpl, weight_buf[na:])
+            if int(weight_ver.value) == ver:  # seqlock: no publish overlapped
+                worker.load_weights(a_sd, c_sd)
+                last_ver = ver
         seqs = worker.rollout(chunk)
         if crash_at and worker.env_steps >= crash_at:
             raise RuntimeError(f"fault injection: actor {actor_id} crash at {crash_at}")
@@ -214,10 +216,15 @@ class ActorPool:
 
     def publish_weights(self, actor_sd: Dict[str, torch.Tensor],
                         critic_sd: Dict[str, torch.Tensor]) -> None:
+        # seqlock: odd version = pack in progress; actors accept only even
+        # versions that are unchanged across their read, so a read that
+        # overlaps a publish is detected and retried (no torn weights)
+        with self.weight_ver.get_lock():
+            self.weight_ver.value += 1  # -> odd
         pack_sd(actor_sd, self.weight_buf[:self._na])
         pack_sd(critic_sd, self.weight_buf[self._na:])
         with self.weight_ver.get_lock():
-            self.weight_ver.value += 1
+            self.weight_ver.value += 1  # -> even (stable)
 
     # ------------------------------------------------- failure detection
     def watchdog(self, stall_seconds: float = 60.0) -> int:

@@ -53,7 +53,10 @@ class GpuSequenceReplay:
         self.per_beta = per_beta
         self.device = torch.device(device)
         self.is_cuda = self.device.type == "cuda"
-        self.fp8_obs = fp8_obs and self.is_cuda
+        # fp8 compresses FLOAT observations 4->1 byte; uint8 pixels are
+        # already 1 byte and e4m3 would both lose precision vs 8-bit and
+        # skip the trunk's /255 normalization — ignore the flag for them
+        self.fp8_obs = fp8_obs and self.is_cuda and obs_dtype != torch.uint8
         L = stored_len
         if self.fp8_obs:
             store_obs_dtype = torch.uint8

@@ -44,8 +44,10 @@ fp8_gather_decode_kernel(const unsigned char* __restrict__ store,
             st_f(&dst[i4 + j], (float)v * inv_scale);
         }
     }
-    // tail
-    if (i4 < row_elems && blockIdx.x == 0 && threadIdx.x == 0) {
+    // tail (row_elems % 4 stragglers): unconditionally thread (0,0) of this
+    // row's blocks — gating it on the thread's own exhausted i4 position
+    // skipped the tail whenever the grid stride jumped past row_elems
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
         for (long i = row_elems & ~3L; i < row_elems; ++i) {
             __hip_fp8_e4m3 v; v.__x = src[i];
             st_f(&dst[i], (float)v * inv_scale);

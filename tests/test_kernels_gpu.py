@@ -221,6 +221,22 @@ def test_fp8_roundtrip_error_bound():
         f"fp8 err {(dec - x).abs().max()} worst at {x[(dec - x).abs().argmax()]}"
 
 
+def test_fp8_gather_odd_row_size_tail():
+    """Row sizes not divisible by 4 (e.g. humanoid 125*67=8375): the gather's
+    vectorized loop leaves a <4-element tail that a buggy guard once skipped,
+    leaving stale memory in the last obs elements of every sampled row."""
+    g = torch.Generator().manual_seed(13)
+    for row in (335, 8375):
+        x = torch.randn(6, row, generator=g).cuda().contiguous()
+        enc = torch.empty_like(x, dtype=torch.uint8)
+        ext().fp8_encode(x.reshape(-1).contiguous(), enc.view(-1), 1.0)
+        idx = torch.tensor([5, 2, 0], dtype=torch.int64).cuda()
+        out = ext().fp8_gather_decode(enc, idx, torch.float32, 1.0)
+        # verify the LAST elements (the tail region) decode correctly
+        for b, s in enumerate([5, 2, 0]):
+            assert torch.allclose(out[b, -5:], x[s, -5:], atol=0.3, rtol=0.08), (row, b)
+
+
 def test_fp8_gather_scrambled_indices():
     g = torch.Generator().manual_seed(11)
     x = torch.randn(8, 64, generator=g).cuda()

@@ -54,9 +54,11 @@ def unpack_sd(template: Dict[str, torch.Tensor], buf: torch.Tensor) -> Dict[str,
 
 
 # -------------------------------------------------------------- actor proc
-def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver,
-                stop_event, heartbeat, crash_at: int, chunk: int = 64,
+def _actor_main(actor_id: int, slot: int, cfg_dict: dict, seq_queue, weight_buf,
+                weight_ver, stop_event, heartbeat, crash_at: int, chunk: int = 64,
                 eval_mode: bool = False, ring=None):
+    # actor_id is GLOBAL (seeds, sigma ladder); slot is the LOCAL pool index
+    # (heartbeat array) — on learner rank r>0 they differ
     torch.set_num_threads(1)
     cfg = Config.from_dict(cfg_dict)
     E = max(cfg.actor.envs_per_actor, 1)
@@ -94,7 +96,7 @@ def _actor_main(actor_id: int, cfg_dict: dict, seq_queue, weight_buf, weight_ver
                 ep_reported = len(worker.episode_returns)
             except queue_mod.Full:
                 pass
-        heartbeat[actor_id] = worker.env_steps
+        heartbeat[slot] = worker.env_steps
 
 
 class ActorPool:
@@ -147,7 +149,7 @@ class ActorPool:
     def _spawn(self, i: int) -> None:
         p = self._ctx.Process(
             target=_actor_main,
-            args=(self.base_actor_id + i, self.cfg.to_dict(), self.seq_queue,
+            args=(self.base_actor_id + i, i, self.cfg.to_dict(), self.seq_queue,
                   self.weight_buf, self.weight_ver, self.stop_event, self.heartbeat,
                   self.crash_at.get(self.base_actor_id + i, 0), self.chunk,
                   i >= self._eval_from, self.rings[i]),

@@ -217,3 +217,32 @@ def test_orchestrator_resume_from_checkpoint(tmp_path):
     cfg.learner.checkpoint_every = 0
     lrn2 = run_training(cfg, device="cpu", resume=ckpt, max_steps=9)
     assert lrn2.step_count == 9  # resumed from 6, ran to 9
+
+
+def test_orchestrated_two_rank_cpu(tmp_path):
+    """The full orchestrated path (actor pool + shm rings + ingest + learner)
+    under torchrun world 2 on gloo — the shape the multi-GPU driver uses.
+    Regression: actors on rank > 0 once crashed on their first heartbeat
+    write (global id indexing a per-rank array) and respawn-looped."""
+    import subprocess
+    import sys
+
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+        "--master-port", "29561", "-m", "pytorch_r2d2_dpg_amd.cli", "train",
+        "--config", "configs/pendulum_cpu.toml",
+        "--set", "actor.num_actors=2", "replay.min_fill=24",
+        "learner.batch_size=8", "learner.log_every=2",
+        "model.hidden_size=32", "model.trunk_sizes=[32]",
+        "replay.seq_len=12", "replay.burn_in=4", "replay.overlap=6",
+        "--steps", "6",
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                       cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    assert "IndexError" not in out and "Traceback" not in out, out[-3000:]
+    step_lines = [l for l in out.splitlines() if "[step " in l]
+    assert step_lines, out[-3000:]
+    assert all("actor_respawns=0" in l for l in step_lines), step_lines

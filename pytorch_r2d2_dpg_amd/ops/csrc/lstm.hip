@@ -388,6 +388,9 @@ static inline int pick_geom(int H, int* threads, int* hf) {
         if (H % cols) continue;
         int waves = H / cols;
         if (waves < 2 || waves > 16) continue;
+        // must fit the kernel's __launch_bounds__ (1024 for HFRAG<=2, else
+        // 512) — a launch wider than the compiled bound fails at dispatch
+        if (waves * 64 > (f <= 2 ? 1024 : 512)) continue;
         *threads = waves * 64;
         *hf = f;
         return 0;

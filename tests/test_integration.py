@@ -229,8 +229,8 @@ def test_orchestrated_two_rank_cpu(tmp_path):
 
     cmd = [
         sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-        "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-        "--master-port", "29561", "-m", "pytorch_r2d2_dpg_amd.cli", "train",
+        "--nproc-per-node", "2", "--standalone", "--local-addr", "127.0.0.1",
+        "-m", "pytorch_r2d2_dpg_amd.cli", "train",
         "--config", "configs/pendulum_cpu.toml",
         "--set", "actor.num_actors=2", "replay.min_fill=24",
         "learner.batch_size=8", "learner.log_every=2",

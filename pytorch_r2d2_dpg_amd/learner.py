@@ -132,6 +132,17 @@ class Learner:
         else:
             self._side_stream = None
             self._ev_ba = self._ev_bc = self._ev_cf = self._ev_ab = None
+        # Paired backward (default on with overlap + bf16; R2D2_NO_PAIR_BWD=1
+        # opts out): the critic-loss backward and the actor-chain critic
+        # backward run as ONE paired (grid.z=2) K2 launch with manual
+        # gradient assembly — two concurrent solo bwd grids contend for CU
+        # residency instead (+3% measured; parameter-equivalent to the
+        # autograd path, test_pair_bwd_step_matches_default).
+        self._pair_bwd = (overlap_ok and self.dtype == torch.bfloat16
+                          and not _os.environ.get("R2D2_NO_PAIR_BWD"))
+        if self._pair_bwd:
+            self._ev_bw = torch.cuda.Event()
+            self._ev_ab2 = torch.cuda.Event()
 
         # optional hipEvent phase timers (SURVEY §5 tracing): R2D2_PHASE_TIMERS=1
         self._phase_events = None
@@ -164,6 +175,8 @@ class Learner:
         unrolls is nearly free parallelism). The CPU path runs the identical
         math on one stream.
         """
+        if getattr(self, "_pair_bwd", False):
+            return self._step_core_pair_bwd(batch, update_priorities, zero_to_none)
         cfg = self.cfg
         Tn = self.train_len + self.n_step
         bi = self.burn_in
@@ -285,6 +298,207 @@ class Learner:
 
         self._pe("actor_opt", 1)
         # ---- priority write-back (stays on-GPU with the K6 tree kernel)
+        if update_priorities and self.replay is not None and batch.get("idx") is not None:
+            self.replay.update_priorities(batch["idx"], prio)
+
+        return {
+            "critic_loss": critic_loss.detach(),
+            "actor_loss": actor_loss.detach(),
+            "mean_q": q_online_bt.detach().mean(),
+            "mean_priority": prio.mean(),
+        }
+
+    # ------------------------------------------------- paired-backward step
+    def _step_core_pair_bwd(self, batch: Dict[str, torch.Tensor],
+                            update_priorities: bool = True,
+                            zero_to_none: bool = True) -> Dict[str, torch.Tensor]:
+        """Experimental gated variant (R2D2_PAIR_BWD=1; GPU bf16 + overlap):
+        identical math to _step_core, but the critic-loss backward and the
+        actor-chain critic backward run as ONE paired (grid.z=2) K2 launch
+        (docs/ROADMAP.md lever 1) with the surrounding dense gradients
+        (gates GEMMs, trunk, heads) assembled manually."""
+        from .ops import hip_ext
+        from .ops.lstm import _FusedLSTMSeq
+        cfg = self.cfg
+        ext = hip_ext()
+        Tn = self.train_len + self.n_step
+        bi = self.burn_in
+        T = self.train_len
+        H = self.actor.hidden_size
+
+        obs = batch["obs"].transpose(0, 1)
+        if obs.dtype != torch.uint8:
+            obs = obs.to(self.dtype)
+        action = batch["action"].transpose(0, 1).to(self.dtype)
+        reward = batch["reward"]
+        done = batch["done"]
+        weight = batch["weight"]
+        st = batch["state"].to(self.dtype)
+        h_a, c_a = st[:, 0].contiguous(), st[:, 1].contiguous()
+        h_c, c_c = st[:, 2].contiguous(), st[:, 3].contiguous()
+
+        main = torch.cuda.current_stream()
+        s1 = self._side_stream
+        s1.wait_stream(main)
+
+        self._pe("burn_in", 0)
+        if bi > 0:
+            with torch.no_grad():
+                feat_ab = self.actor.trunk(obs[:bi])
+                feat_cb = torch.cat([self.critic.trunk(obs[:bi]), action[:bi]], dim=-1)
+                la, lcb = self.actor.lstm, self.critic.lstm
+                _, h_a2, c_a2, _, h_c2, c_c2 = lstm_seq_pair(
+                    feat_ab, h_a, c_a, la.weight_x, la.weight_h, la.bias,
+                    feat_cb, h_c, c_c, lcb.weight_x, lcb.weight_h, lcb.bias)
+        else:
+            h_a2, c_a2, h_c2, c_c2 = h_a, c_a, h_c, c_c
+        self._ev_ba.record(main)
+
+        lc = self.critic.lstm
+        wx_c, wh_c, b_c = lc.weight_x, lc.weight_h, lc.bias
+        obs_w = obs[bi:bi + T]
+        c_c2d = c_c2.detach().contiguous()
+
+        # ---- side stream: actor chain FORWARD only + the q_pi head seed.
+        # The critic unroll here is a RAW stash forward (its weight grads
+        # are discarded by design — pre-update-critic DDPG), so only dmu is
+        # needed from its backward, which the paired K2 below provides.
+        with torch.cuda.stream(s1):
+            s1.wait_event(self._ev_ba)
+            la = self.actor.lstm
+            feat_a = self.actor.trunk(obs_w)
+            hs_a, _ = _FusedLSTMSeq.apply(feat_a, h_a2.detach(), c_a2.detach(),
+                                          la.weight_x, la.weight_h, la.bias)
+            mu = (torch.tanh(self.actor.head(hs_a.reshape(-1, H)))
+                  .view(T, -1, self.actor.action_dim) * self.actor.action_scale)
+            with torch.no_grad():
+                feat_qc = self.critic.trunk(obs_w)
+                x_q = torch.cat([feat_qc, mu.detach()], dim=-1)
+                gx_q = torch.addmm(b_c, x_q.reshape(T * x_q.shape[1], -1),
+                                   wx_c.t()).view(T, -1, 4 * H).contiguous()
+                hs_q, _, ga_q, cs_q = ext.lstm_seq_fwd(
+                    gx_q, h_c2.detach().contiguous(), c_c2d, wh_c.contiguous())
+            hs_q_leaf = hs_q.detach().requires_grad_(True)
+            q_pi = self.critic.head(hs_q_leaf.reshape(-1, H)).view(T, -1)
+            actor_loss = -q_pi.float().mean()
+            dhs_q, = torch.autograd.grad(actor_loss, [hs_q_leaf])
+            self._ev_ab.record(s1)
+
+        self._pe("burn_in", 1)
+        self._pe("target", 0)
+        # ---- main: TA + CF (raw stash) paired fwd, TC, K3, critic loss
+        ta = self.target_actor
+        feat_cf = self.critic.trunk(obs_w)            # grad (trunk params)
+        x_cf = torch.cat([feat_cf, action[bi:bi + T]], dim=-1)
+        with torch.no_grad():
+            feat_ta = ta.trunk(obs[bi:bi + Tn])
+            gx_ta = torch.addmm(ta.lstm.bias,
+                                feat_ta.reshape(Tn * feat_ta.shape[1], -1),
+                                ta.lstm.weight_x.t()).view(Tn, -1, 4 * H).contiguous()
+            gx_cf = torch.addmm(b_c, x_cf.detach().reshape(T * x_cf.shape[1], -1),
+                                wx_c.t()).view(T, -1, 4 * H).contiguous()
+            hs_ta, _, _, _, hs_cf, _, ga_cf, cs_cf = ext.lstm_seq_fwd_pair(
+                gx_ta, h_a2.contiguous(), c_a2.contiguous(),
+                ta.lstm.weight_h.contiguous(),
+                gx_cf, h_c2.detach().contiguous(), c_c2d, wh_c.contiguous(),
+                False, True)
+            mu_t = (torch.tanh(ta.head(hs_ta.reshape(-1, H)))
+                    .view(Tn, -1, ta.action_dim) * ta.action_scale)
+            qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
+            q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
+        hs_cf_leaf = hs_cf.detach().requires_grad_(True)
+        q_online = self.critic.head(hs_cf_leaf.reshape(-1, H)).view(T, -1)
+        q_online_bt = q_online.transpose(0, 1).float()
+        with torch.no_grad():
+            y, td_abs, prio = td_targets_and_priorities(
+                reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),
+                self.gamma, self.n_step, self.eta,
+                rescale=cfg.learner.value_rescale,
+            )
+        self._pe("target", 1)
+        self._pe("critic", 0)
+        critic_loss = (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
+        head_p = [self.critic.head.weight, self.critic.head.bias]
+        dhs_cf, dw_head, db_head = torch.autograd.grad(
+            critic_loss, [hs_cf_leaf] + head_p)
+
+        # ---- joint paired K2: both critic-LSTM backwards in one launch
+        main.wait_event(self._ev_ab)
+        for t_ in (dhs_q, ga_q, cs_q):
+            t_.record_stream(main)
+        zero = torch.zeros_like(h_c2)
+        whT_c = wh_c.t().contiguous()
+        dg_cf, _, _, dg_q, _, _ = ext.lstm_seq_bwd_pair(
+            dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c,
+            dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
+        self._ev_bw.record(main)
+
+        # ---- side: finish the actor chain (dmu -> actor grads), overlapping
+        # the critic dense-gradient GEMMs below
+        dg_q.record_stream(s1)
+        with torch.cuda.stream(s1):
+            s1.wait_event(self._ev_bw)
+            F_q = feat_qc.shape[-1]
+            dx_q = (dg_q.reshape(-1, 4 * H) @ wx_c).view(T, -1, x_q.shape[-1])
+            dmu = dx_q[..., F_q:].contiguous()
+            actor_grads = torch.autograd.grad(mu, self._actor_params,
+                                              grad_outputs=dmu)
+            self._ev_ab2.record(s1)
+
+        # ---- main: critic dense grads (same math as _FusedLSTMSeq.backward)
+        B_ = x_cf.shape[1]
+        dgf = dg_cf.reshape(T * B_, 4 * H)
+        dwx_c = dgf.t() @ x_cf.detach().reshape(T * B_, -1)
+        h_prev = torch.cat([h_c2.detach().unsqueeze(0), hs_cf[:-1]],
+                           dim=0).reshape(T * B_, H)
+        dwh_c = dgf.t() @ h_prev
+        db_c = dgf.float().sum(dim=0).to(dgf.dtype)
+        dx_cf = (dgf @ wx_c).view(T, B_, -1)
+        F_cf = feat_cf.shape[-1]
+        trunk_params = list(self.critic.trunk.parameters())
+        trunk_grads = torch.autograd.grad(
+            feat_cf, trunk_params, grad_outputs=dx_cf[..., :F_cf].contiguous()
+        ) if trunk_params else []
+        gmap = {id(p): g for p, g in zip(trunk_params, trunk_grads)}
+        gmap[id(wx_c)] = dwx_c
+        gmap[id(wh_c)] = dwh_c
+        gmap[id(b_c)] = db_c
+        gmap[id(head_p[0])] = dw_head
+        gmap[id(head_p[1])] = db_head
+        cparams = list(self.critic.parameters())
+        self.opt_critic.zero_grad(set_to_none=zero_to_none)
+        for p in cparams:
+            g = gmap.get(id(p))
+            assert g is not None, "unmapped critic parameter in pair-bwd path"
+            if p.grad is None:
+                p.grad = g if zero_to_none else g.clone()
+            else:
+                p.grad.copy_(g)
+        self.dist.allreduce_grads(cparams, key="critic")
+        if cfg.learner.max_grad_norm > 0:
+            torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
+        # the side stream reads critic params (head fwd, wx_c): join before write
+        main.wait_event(self._ev_ab2)
+        self.opt_critic.step()
+
+        self._pe("critic", 1)
+        self._pe("actor_opt", 0)
+        for p, g in zip(self._actor_params, actor_grads):
+            if p.grad is None:
+                p.grad = g.clone() if not zero_to_none else g
+            else:
+                p.grad.copy_(g)
+        self.dist.allreduce_grads(self._actor_params, key="actor")
+        if cfg.learner.max_grad_norm > 0:
+            torch.nn.utils.clip_grad_norm_(self._actor_params,
+                                           cfg.learner.max_grad_norm)
+        self.opt_actor.step()
+
+        if self.step_count % cfg.learner.target_update_every == 0:
+            self.polyak_actor(cfg.learner.tau)
+            self.polyak_critic(cfg.learner.tau)
+
+        self._pe("actor_opt", 1)
         if update_priorities and self.replay is not None and batch.get("idx") is not None:
             self.replay.update_priorities(batch["idx"], prio)
 

@@ -50,6 +50,12 @@ extern "C" int launch_lstm_bwd_px_bf16(const void*, const void*, const void*, co
                                        const float*, const void*, const void*, void*, void*,
                                        void*, unsigned char*, unsigned long long*,
                                        unsigned int*, int, int, int, hipStream_t);
+extern "C" int launch_lstm_bwd_px2_bf16(
+    const void*, const void*, const void*, const void*, const float*, const void*,
+    const void*, void*, void*, void*, unsigned char*, unsigned long long*, int,
+    const void*, const void*, const void*, const void*, const float*, const void*,
+    const void*, void*, void*, void*, unsigned char*, unsigned long long*, int,
+    unsigned int*, int, int, hipStream_t);
 
 namespace {
 
@@ -291,6 +297,58 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
     return {dgates, dh0, dc0};
 }
 
+// ---- K2-pair: two independent LSTM backwards in ONE launch ----------------
+// Same B/H; T may differ. Each set gets its own slab/flags workspace. Used
+// for the critic-loss backward + the actor-chain critic backward, which
+// become ready at nearly the same point of the learner step.
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
+           torch::Tensor, torch::Tensor, torch::Tensor>
+lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
+                  torch::Tensor ga0, torch::Tensor cs0, torch::Tensor c00,
+                  torch::Tensor whT0,
+                  torch::Tensor dh1s, torch::Tensor dhT1, torch::Tensor dcT1,
+                  torch::Tensor ga1, torch::Tensor cs1, torch::Tensor c01,
+                  torch::Tensor whT1) {
+    for (auto* t : {&dh0s, &dhT0, &dcT0, &ga0, &c00, &whT0,
+                    &dh1s, &dhT1, &dcT1, &ga1, &c01, &whT1})
+        check_dev_contig(*t, "bwd pair arg");
+    const auto T0 = dh0s.size(0);
+    const auto T1 = dh1s.size(0);
+    const auto B = dh0s.size(1);
+    const auto H = dh0s.size(2);
+    TORCH_CHECK(dh1s.size(1) == B && dh1s.size(2) == H, "bwd pair shape mismatch");
+    TORCH_CHECK(dh0s.scalar_type() == torch::kBFloat16, "bwd pair is bf16-only");
+    auto wp0 = pack_fragment_major(whT0);
+    auto wp1 = pack_fragment_major(whT1);
+    const int RT = (int)((B + 15) / 16);
+    auto opts = dh0s.options();
+    auto dg0 = torch::empty({T0, B, 4 * H}, opts);
+    auto dg1 = torch::empty({T1, B, 4 * H}, opts);
+    auto dh00 = torch::empty({B, H}, opts);
+    auto dh01 = torch::empty({B, H}, opts);
+    auto dc00 = torch::empty({B, H}, opts);
+    auto dc01 = torch::empty({B, H}, opts);
+    auto slab0 = torch::empty({2LL, RT, 16 * 4 * H * 2}, opts.dtype(torch::kUInt8));
+    auto slab1 = torch::empty({2LL, RT, 16 * 4 * H * 2}, opts.dtype(torch::kUInt8));
+    auto flags0 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
+    auto flags1 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
+    auto& err = px_err_word(dh0s.device());
+    int rc = launch_lstm_bwd_px2_bf16(
+        dh0s.data_ptr(), dhT0.data_ptr(), dcT0.data_ptr(), ga0.data_ptr(),
+        cs0.data_ptr<float>(), c00.data_ptr(), wp0.data_ptr(),
+        dg0.data_ptr(), dh00.data_ptr(), dc00.data_ptr(),
+        slab0.data_ptr<uint8_t>(), (unsigned long long*)flags0.data_ptr<int64_t>(),
+        (int)T0,
+        dh1s.data_ptr(), dhT1.data_ptr(), dcT1.data_ptr(), ga1.data_ptr(),
+        cs1.data_ptr<float>(), c01.data_ptr(), wp1.data_ptr(),
+        dg1.data_ptr(), dh01.data_ptr(), dc01.data_ptr(),
+        slab1.data_ptr<uint8_t>(), (unsigned long long*)flags1.data_ptr<int64_t>(),
+        (int)T1,
+        (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, cur_stream());
+    TORCH_CHECK(rc == 0, "bwd pair geometry unsupported (fall back to two calls)");
+    return {dg0, dh00, dc00, dg1, dh01, dc01};
+}
+
 // ---- K3: fused n-step TD targets + priorities -----------------------------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
 td_targets(torch::Tensor rewards, torch::Tensor dones, torch::Tensor q_boot,
@@ -411,6 +469,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("gx0"), py::arg("h00"), py::arg("c00"), py::arg("wh0"),
           py::arg("gx1"), py::arg("h01"), py::arg("c01"), py::arg("wh1"),
           py::arg("stash0") = false, py::arg("stash1") = false);
+    m.def("lstm_seq_bwd_pair", &lstm_seq_bwd_pair,
+          "two independent LSTM backwards in one launch (per-set T)");
     m.def("set_px_serialize", &set_px_serialize,
           "serialize column-split kernels device-wide (off by default)");
 }

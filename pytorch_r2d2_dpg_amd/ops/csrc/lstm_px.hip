@@ -277,23 +277,40 @@ extern "C" int launch_lstm_fwd_px_bf16(
 // step (flag stored only after every wave drained its sc1 stores — G16 R1),
 // sweeps peers' slabs into the LDS tile, then runs the dh recurrence MFMA
 // with K = 4H against its whT row slice (2MB/CG streamed per step).
+// Per-set backward arguments — one set for the solo kernel, two for the
+// paired (grid.z=2) kernel. Every set has its OWN slab/flags workspace.
+struct BwdPxArgs {
+    const __hip_bfloat16* dh_seq;    // [T,B,H]
+    const __hip_bfloat16* dhT;       // [B,H]
+    const __hip_bfloat16* dcT;       // [B,H]
+    const __hip_bfloat16* gates_act; // [T,B,4H]
+    const float* c_seq;              // [T,B,H]
+    const __hip_bfloat16* c0;        // [B,H]
+    const __hip_bfloat16* whT;       // packed [H/16][4H/32][64][8]
+    __hip_bfloat16* dgates;          // [T,B,4H]
+    __hip_bfloat16* dh0;             // [B,H]
+    __hip_bfloat16* dc0;             // [B,H]
+    unsigned char* slab;             // [2][RT][16*4H*2] bytes
+    unsigned long long* flags;       // [2][RT][CG] zeroed
+    int T;
+};
+
 template <int CG>
-__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
-lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
-                   const __hip_bfloat16* __restrict__ dhT,      // [B,H]
-                   const __hip_bfloat16* __restrict__ dcT,      // [B,H]
-                   const __hip_bfloat16* __restrict__ gates_act,// [T,B,4H]
-                   const float* __restrict__ c_seq,             // [T,B,H]
-                   const __hip_bfloat16* __restrict__ c0,       // [B,H]
-                   const __hip_bfloat16* __restrict__ whT,      // packed [H/16][4H/32][64][8]
-                   __hip_bfloat16* __restrict__ dgates,         // [T,B,4H]
-                   __hip_bfloat16* __restrict__ dh0,            // [B,H]
-                   __hip_bfloat16* __restrict__ dc0,            // [B,H]
-                   unsigned char* __restrict__ slab,            // [2][RT][16*4H*2] bytes
-                   unsigned long long* __restrict__ flags,      // [2][RT][CG] zeroed
-                   unsigned int* __restrict__ err,
-                   int T, int B, int H) {
-    extern __shared__ __attribute__((aligned(16))) char smem[];  // dg tile [16][4H]
+DEV_INLINE void lstm_bwd_px_body(const BwdPxArgs& a, unsigned int* err,
+                                 int B, int H, char* smem) {
+    const __hip_bfloat16* __restrict__ dh_seq = a.dh_seq;
+    const __hip_bfloat16* __restrict__ dhT = a.dhT;
+    const __hip_bfloat16* __restrict__ dcT = a.dcT;
+    const __hip_bfloat16* __restrict__ gates_act = a.gates_act;
+    const float* __restrict__ c_seq = a.c_seq;
+    const __hip_bfloat16* __restrict__ c0 = a.c0;
+    const __hip_bfloat16* __restrict__ whT = a.whT;
+    __hip_bfloat16* __restrict__ dgates = a.dgates;
+    __hip_bfloat16* __restrict__ dh0 = a.dh0;
+    __hip_bfloat16* __restrict__ dc0 = a.dc0;
+    unsigned char* __restrict__ slab = a.slab;
+    unsigned long long* __restrict__ flags = a.flags;
+    const int T = a.T;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -489,6 +506,26 @@ lstm_bwd_px_kernel(const __hip_bfloat16* __restrict__ dh_seq,   // [T,B,H]
     }
 }
 
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_bwd_px_kernel(BwdPxArgs a, unsigned int* err, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];  // dg tile [16][4H]
+    lstm_bwd_px_body<CG>(a, err, B, H, smem);
+}
+
+// Paired (grid.z=2) backward: two independent unrolls (same B/H, T may
+// differ), each with its own slab/flags workspace — e.g. the critic-loss
+// backward and the actor-chain critic backward, which become ready at
+// nearly the same point in the step. One uniform grid co-dispatches both
+// (RT*CG*2 blocks), avoiding the cross-stream CU contention of two
+// concurrent solo bwd grids.
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_bwd_px2_kernel(BwdPxArgs a0, BwdPxArgs a1, unsigned int* err, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    lstm_bwd_px_body<CG>(blockIdx.z ? a1 : a0, err, B, H, smem);
+}
+
 extern "C" int launch_lstm_bwd_px_bf16(
     const void* dh_seq, const void* dhT, const void* dcT, const void* gates_act,
     const float* c_seq, const void* c0, const void* whT_packed,
@@ -515,20 +552,61 @@ extern "C" int launch_lstm_bwd_px_bf16(
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG);
     size_t lds = (size_t)16 * 4 * H * 2;
-#define BWD_PX_CALL(CGV)                                                        \
-    lstm_bwd_px_kernel<CGV><<<grid, dim3(waves * 64), lds, stream>>>(           \
-        (const __hip_bfloat16*)dh_seq, (const __hip_bfloat16*)dhT,              \
-        (const __hip_bfloat16*)dcT, (const __hip_bfloat16*)gates_act, c_seq,    \
-        (const __hip_bfloat16*)c0, (const __hip_bfloat16*)whT_packed,           \
-        (__hip_bfloat16*)dgates, (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0,    \
-        slab, flags, err, T, B, H)
+    BwdPxArgs a{(const __hip_bfloat16*)dh_seq, (const __hip_bfloat16*)dhT,
+                (const __hip_bfloat16*)dcT, (const __hip_bfloat16*)gates_act,
+                c_seq, (const __hip_bfloat16*)c0,
+                (const __hip_bfloat16*)whT_packed, (__hip_bfloat16*)dgates,
+                (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0, slab, flags, T};
     switch (CG) {
-        case 2: BWD_PX_CALL(2); break;
-        case 4: BWD_PX_CALL(4); break;
-        case 8: BWD_PX_CALL(8); break;
+        case 2: lstm_bwd_px_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+        case 4: lstm_bwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+        case 8: lstm_bwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
         default: return -1;
     }
-#undef BWD_PX_CALL
+    return 0;
+}
+
+extern "C" int launch_lstm_bwd_px2_bf16(
+    const void* dh_seq0, const void* dhT0, const void* dcT0, const void* ga0,
+    const float* cs0, const void* c00, const void* whT0, void* dg0, void* dh00,
+    void* dc00, unsigned char* slab0, unsigned long long* flags0, int T0,
+    const void* dh_seq1, const void* dhT1, const void* dcT1, const void* ga1,
+    const float* cs1, const void* c01, const void* whT1, void* dg1, void* dh01,
+    void* dc01, unsigned char* slab1, unsigned long long* flags1, int T1,
+    unsigned int* err, int B, int H, hipStream_t stream) {
+    int RT = (B + 15) / 16;
+    if (H & (H - 1)) return -1;
+    if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;
+    int CG = 0;
+    for (int c : {8, 4, 2}) {
+        if (H % (16 * c)) continue;
+        int w = (H / c) / 16;
+        if (w < 2 || w > 16) continue;
+        if (H / c < 64) continue;        // slab swizzle closure (see solo)
+        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
+        CG = c;
+        break;
+    }
+    if (CG == 0) return -1;
+    int waves = (H / CG) / 16;
+    dim3 grid(RT, CG, 2);
+    size_t lds = (size_t)16 * 4 * H * 2;
+    BwdPxArgs a0{(const __hip_bfloat16*)dh_seq0, (const __hip_bfloat16*)dhT0,
+                 (const __hip_bfloat16*)dcT0, (const __hip_bfloat16*)ga0, cs0,
+                 (const __hip_bfloat16*)c00, (const __hip_bfloat16*)whT0,
+                 (__hip_bfloat16*)dg0, (__hip_bfloat16*)dh00,
+                 (__hip_bfloat16*)dc00, slab0, flags0, T0};
+    BwdPxArgs a1{(const __hip_bfloat16*)dh_seq1, (const __hip_bfloat16*)dhT1,
+                 (const __hip_bfloat16*)dcT1, (const __hip_bfloat16*)ga1, cs1,
+                 (const __hip_bfloat16*)c01, (const __hip_bfloat16*)whT1,
+                 (__hip_bfloat16*)dg1, (__hip_bfloat16*)dh01,
+                 (__hip_bfloat16*)dc01, slab1, flags1, T1};
+    switch (CG) {
+        case 2: lstm_bwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+        case 4: lstm_bwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+        case 8: lstm_bwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+        default: return -1;
+    }
     return 0;
 }
 

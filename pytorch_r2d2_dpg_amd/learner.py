@@ -125,13 +125,11 @@ class Learner:
                       and not _os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()
-            self._ev_ba = torch.cuda.Event()
-            self._ev_bc = torch.cuda.Event()
-            self._ev_cf = torch.cuda.Event()
-            self._ev_ab = torch.cuda.Event()
+            self._ev_ba = torch.cuda.Event()   # burn-in done (main -> side)
+            self._ev_ab = torch.cuda.Event()   # side chain done (side -> main)
         else:
             self._side_stream = None
-            self._ev_ba = self._ev_bc = self._ev_cf = self._ev_ab = None
+            self._ev_ba = self._ev_ab = None
         # Paired backward (default on with overlap + bf16; R2D2_NO_PAIR_BWD=1
         # opts out): the critic-loss backward and the actor-chain critic
         # backward run as ONE paired (grid.z=2) K2 launch with manual
@@ -196,8 +194,6 @@ class Learner:
 
         s1 = self._side_stream
         ev_ba = self._ev_ba
-        ev_bc = self._ev_bc
-        ev_cf = self._ev_cf
         ev_ab = self._ev_ab
 
         main = torch.cuda.current_stream() if overlap else None

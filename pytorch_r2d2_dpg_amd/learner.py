@@ -11,10 +11,12 @@ are all-reduced as one fused bucket over RCCL/xGMI after each backward
 (parallel/dist.py) before the Adam step — DP ranks therefore hold identical
 weights at all times; rank 0 alone checkpoints/logs.
 
-Execution (GPU): the burn-in pair runs as ONE column-split launch
-(grid.z=2); the actor chain (through the PRE-update critic) runs on a side
-HIP stream overlapped with the target/critic chain at B<=256; the two
-gradient all-reduces are issued on the main stream in a fixed order.
+Execution (GPU): independent unrolls share paired (grid.z=2) column-split
+launches — the two burn-ins, the target-actor unroll with the online
+critic forward, and (bf16 + overlap) the two critic-LSTM backwards; the
+actor chain (through the PRE-update critic) runs on a side HIP stream
+overlapped with the target/critic chain at B<=256 and train_len<=48; the
+two gradient all-reduces are issued on the main stream in a fixed order.
 ``capture_graph()`` can record the whole update into a hipGraph, but eager
 + stream overlap measures faster at these sizes (steady-state device
 occupancy is already ~92%), so it is not the default.
@@ -308,11 +310,13 @@ class Learner:
     def _step_core_pair_bwd(self, batch: Dict[str, torch.Tensor],
                             update_priorities: bool = True,
                             zero_to_none: bool = True) -> Dict[str, torch.Tensor]:
-        """Experimental gated variant (R2D2_PAIR_BWD=1; GPU bf16 + overlap):
-        identical math to _step_core, but the critic-loss backward and the
+        """Default GPU step at bf16 + overlap (R2D2_NO_PAIR_BWD=1 falls back
+        to _step_core): identical math, but the critic-loss backward and the
         actor-chain critic backward run as ONE paired (grid.z=2) K2 launch
-        (docs/ROADMAP.md lever 1) with the surrounding dense gradients
-        (gates GEMMs, trunk, heads) assembled manually."""
+        with the surrounding dense gradients (gates GEMMs, trunk, heads)
+        assembled manually — two concurrent solo bwd grids contend for CU
+        residency instead. Parameter-equivalence is locked by
+        test_pair_bwd_step_matches_default."""
         from .ops import hip_ext
         from .ops.lstm import _FusedLSTMSeq
         cfg = self.cfg

@@ -81,8 +81,9 @@ def build(force: bool = False, verbose: bool = False) -> str:
     libdirs = [f"-L{p}" for p in ce.library_paths("cuda")]
     libs = ["-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",
             "-ltorch_hip", "-lc10_hip", "-lamdhip64"]
-    if force or any(os.path.getmtime(o) > os.path.getmtime(OUT)
-                    for o in objs) if os.path.exists(OUT) else True:
+    need_link = (not os.path.exists(OUT) or force
+                 or any(os.path.getmtime(o) > os.path.getmtime(OUT) for o in objs))
+    if need_link:
         _run([hipcc, "-shared", "-fPIC", *objs, *libdirs, *libs, "-o", OUT])
     return OUT
 

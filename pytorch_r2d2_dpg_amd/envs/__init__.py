@@ -20,6 +20,18 @@ def make_env(name: str, seed: int = 0, **kw) -> Env:
         return PendulumEnv(seed=seed, **kw)
     if name in SYNTHETIC_SPECS:
         return SyntheticEnv(name, seed=seed, **kw)
+    # "domain-task" names outside the synthetic set load real DM-Control
+    # when the (optional) dependency is installed (envs/dmc.py)
+    if "-" in name:
+        domain, task = name.split("-", 1)
+        try:
+            from .dmc import DMControlEnv
+            return DMControlEnv(domain, task, seed=seed, **kw)
+        except ImportError:
+            raise ValueError(
+                f"env '{name}': dm_control is not installed in this "
+                f"environment (offline sandbox); available: pendulum, "
+                f"{', '.join(SYNTHETIC_SPECS)}") from None
     raise ValueError(f"unknown env '{name}' (have: pendulum, {', '.join(SYNTHETIC_SPECS)})")
 
 

@@ -78,3 +78,73 @@ def test_max_env_steps_caps_rollout():
     more = w.env_steps
     w.rollout(30)
     assert w.env_steps == more  # capped: no further progress
+
+
+def test_dmc_adapter_with_stubbed_suite(monkeypatch):
+    """The optional DM-Control adapter: flattening, spec derivation and the
+    termination/truncation split from ts.discount (exercised against a stub
+    suite — the real dm_control is absent in this sandbox)."""
+    import sys
+    import types
+
+    class _TS:
+        def __init__(self, obs, reward, discount, last):
+            self.observation = obs
+            self.reward = reward
+            self.discount = discount
+            self._last = last
+
+        def last(self):
+            return self._last
+
+    class _ASpec:
+        shape = (2,)
+        maximum = np.array([1.0, 1.0])
+
+    class _FakeEnv:
+        def __init__(self):
+            self.t = 0
+
+        def action_spec(self):
+            return _ASpec()
+
+        def observation_spec(self):
+            class _S:  # shaped entries: (2,) vector + scalar
+                pass
+            v = _S(); v.shape = (2,)
+            s = _S(); s.shape = ()
+            return {"pos": v, "height": s}
+
+        def reset(self):
+            self.t = 0
+            return _TS({"pos": np.array([1.0, 2.0]), "height": 3.0}, None, 1.0, False)
+
+        def step(self, a):
+            self.t += 1
+            if self.t >= 3:  # terminate (discount 0)
+                return _TS({"pos": np.zeros(2), "height": 0.0}, 1.0, 0.0, True)
+            return _TS({"pos": np.ones(2), "height": 1.0}, 0.5, 1.0, False)
+
+    suite = types.ModuleType("dm_control.suite")
+    suite.load = lambda d, t, task_kwargs=None: _FakeEnv()
+    pkg = types.ModuleType("dm_control")
+    pkg.suite = suite
+    monkeypatch.setitem(sys.modules, "dm_control", pkg)
+    monkeypatch.setitem(sys.modules, "dm_control.suite", suite)
+
+    env = make_env("walker-walk", seed=3)
+    assert env.spec.obs_shape == (3,) and env.spec.action_dim == 2
+    obs = env.reset()
+    # sorted key order: height then pos
+    assert np.allclose(obs, [3.0, 1.0, 2.0])
+    obs, r, done, trunc = env.step(np.zeros(2, np.float32))
+    assert r == 0.5 and not done and not trunc
+    env.step(np.zeros(2, np.float32))
+    obs, r, done, trunc = env.step(np.zeros(2, np.float32))
+    assert done and r == 1.0  # discount 0 at last() -> true termination
+
+
+def test_make_env_unknown_dmc_raises_clear_error():
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match="dm_control is not installed"):
+        make_env("walker-walk")

@@ -20,7 +20,7 @@ from .config import Config
 from .envs import make_env
 from .models import make_models
 from .replay.sequence import Sequence, SequenceCutter
-from .utils.noise import GaussianNoise, sigma_ladder
+from .utils.noise import make_noise, sigma_ladder
 
 
 class ActorWorker:
@@ -40,7 +40,8 @@ class ActorWorker:
             sigma = 0.0 if eval_mode else sigma_ladder(
                 actor_id, max(cfg.actor.num_actors, 1), cfg.actor.sigma_max, cfg.actor.sigma_alpha)
         self.sigma = sigma
-        self.noise = GaussianNoise(spec.action_dim, sigma, seed=cfg.env.seed + actor_id + 7777)
+        self.noise = make_noise(cfg.actor.noise, spec.action_dim, sigma,
+                                seed=cfg.env.seed + actor_id + 7777)
         R = cfg.replay
         self.cutter = SequenceCutter(R.seq_len, R.overlap, R.burn_in, R.n_step,
                                      R.gamma, R.priority_eta, cfg.model.hidden_size)
@@ -101,6 +102,7 @@ class ActorWorker:
                 self.episode_returns.append(self.episode_return)
                 self.episode_return = 0.0
                 self._obs = self.env.reset()
+                self.noise.reset()
                 H = self.cfg.model.hidden_size
                 h_a = torch.zeros(1, H); c_a = torch.zeros(1, H)
                 h_c = torch.zeros(1, H); c_c = torch.zeros(1, H)
@@ -136,8 +138,8 @@ class VectorActorWorker:
         base_sigma = 0.0 if eval_mode else sigma_ladder(
             actor_id, max(cfg.actor.num_actors, 1), cfg.actor.sigma_max,
             cfg.actor.sigma_alpha)
-        self.noise = [GaussianNoise(spec.action_dim, base_sigma,
-                                    seed=cfg.env.seed + actor_id * 131 + e + 7777)
+        self.noise = [make_noise(cfg.actor.noise, spec.action_dim, base_sigma,
+                                 seed=cfg.env.seed + actor_id * 131 + e + 7777)
                       for e in range(num_envs)]
         R = cfg.replay
         self.cutters = [SequenceCutter(R.seq_len, R.overlap, R.burn_in, R.n_step,
@@ -198,6 +200,7 @@ class VectorActorWorker:
                     self.episode_returns.append(float(self.episode_return[e]))
                     self.episode_return[e] = 0.0
                     self._obs[e] = env.reset()
+                    self.noise[e].reset()
                     for t in (self._h_a, self._c_a, self._h_c, self._c_c):
                         t[e].zero_()
             self.env_steps += E

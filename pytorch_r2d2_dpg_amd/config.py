@@ -78,6 +78,7 @@ class LearnerConfig:
 class ActorConfig:
     num_actors: int = 4
     envs_per_actor: int = 1         # vectorized actor: envs batched per process
+    noise: str = "gaussian"         # gaussian | ou (Ornstein-Uhlenbeck, DDPG paper)
     sigma_max: float = 0.4          # Ape-X ladder: sigma_i = sigma_max^(1 + alpha*i/(N-1))
     sigma_alpha: float = 7.0
     eval_actors: int = 0            # noise-free actors for evaluation logging

@@ -26,3 +26,38 @@ class GaussianNoise:
 
     def __call__(self) -> np.ndarray:
         return self._rng.standard_normal(self.action_dim).astype(np.float32) * self.sigma
+
+    def reset(self) -> None:
+        pass
+
+
+class OUNoise:
+    """Ornstein-Uhlenbeck process (the DDPG paper's temporally-correlated
+    exploration): dx = theta*(mu - x)*dt + sigma*sqrt(dt)*N(0,1)."""
+
+    def __init__(self, action_dim: int, sigma: float, seed: int = 0,
+                 theta: float = 0.15, dt: float = 1e-2, mu: float = 0.0):
+        self.sigma = sigma
+        self.theta = theta
+        self.dt = dt
+        self.mu = mu
+        self.action_dim = action_dim
+        self._rng = np.random.default_rng(seed)
+        self._x = np.full(action_dim, mu, dtype=np.float32)
+
+    def __call__(self) -> np.ndarray:
+        self._x = (self._x + self.theta * (self.mu - self._x) * self.dt
+                   + self.sigma * np.sqrt(self.dt)
+                   * self._rng.standard_normal(self.action_dim).astype(np.float32))
+        return self._x.astype(np.float32)
+
+    def reset(self) -> None:
+        self._x = np.full(self.action_dim, self.mu, dtype=np.float32)
+
+
+def make_noise(kind: str, action_dim: int, sigma: float, seed: int = 0):
+    if kind == "ou":
+        return OUNoise(action_dim, sigma, seed=seed)
+    if kind in ("gaussian", "normal"):
+        return GaussianNoise(action_dim, sigma, seed=seed)
+    raise ValueError(f"unknown noise kind '{kind}' (gaussian | ou)")

@@ -148,3 +148,34 @@ def test_make_env_unknown_dmc_raises_clear_error():
     import pytest as _pytest
     with _pytest.raises(ValueError, match="dm_control is not installed"):
         make_env("walker-walk")
+
+
+def test_ou_noise_temporal_correlation_and_reset():
+    from pytorch_r2d2_dpg_amd.utils.noise import OUNoise, make_noise
+    n = OUNoise(4, sigma=0.3, seed=0)
+    xs = np.stack([n() for _ in range(200)])
+    # OU is temporally correlated: successive-step correlation well above iid
+    a, b = xs[:-1].reshape(-1), xs[1:].reshape(-1)
+    corr = np.corrcoef(a, b)[0, 1]
+    assert corr > 0.8
+    n.reset()
+    assert np.allclose(n._x, 0.0)
+    assert isinstance(make_noise("ou", 2, 0.1), OUNoise)
+
+
+def test_actor_with_ou_noise_rolls_out():
+    from pytorch_r2d2_dpg_amd.actor import ActorWorker
+    from pytorch_r2d2_dpg_amd.config import Config
+
+    cfg = Config()
+    cfg.env.name = "pendulum"
+    cfg.actor.noise = "ou"
+    cfg.model.hidden_size = 16
+    cfg.model.trunk_sizes = (16,)
+    cfg.replay.seq_len = 8
+    cfg.replay.burn_in = 2
+    cfg.replay.overlap = 4
+    cfg.replay.n_step = 2
+    w = ActorWorker(cfg, actor_id=0, sigma=0.3)
+    seqs = w.rollout(30)
+    assert w.env_steps == 30 and seqs

@@ -518,6 +518,10 @@ class Learner:
         """
         assert self.device.type == "cuda", "graph capture needs a GPU"
         assert self.cfg.learner.target_update_every == 1
+        # the paired-backward path uses record_stream (allocator hints),
+        # which is not legal inside graph capture — capture the standard
+        # autograd path instead (identical math)
+        self._pair_bwd = False
         keys = ("obs", "action", "reward", "done", "state", "weight", "idx")
         self._static = {k: example_batch[k].clone() for k in keys}
         # capturable Adam: state must live on device, step as device tensor

@@ -33,10 +33,16 @@ class GaussianNoise:
 
 class OUNoise:
     """Ornstein-Uhlenbeck process (the DDPG paper's temporally-correlated
-    exploration): dx = theta*(mu - x)*dt + sigma*sqrt(dt)*N(0,1)."""
+    exploration): dx = theta*(mu - x)*dt + sigma*sqrt(dt)*N(0,1).
+
+    dt defaults to 1 (the DDPG-paper implementation convention: one OU step
+    per env step), giving stationary std ~ sigma/sqrt(2*theta - theta^2)
+    ~ 1.9*sigma at theta=0.15 — with dt=1e-2 the stationary std collapses
+    to ~0.18*sigma and exploration is too weak to learn (measured on
+    Pendulum)."""
 
     def __init__(self, action_dim: int, sigma: float, seed: int = 0,
-                 theta: float = 0.15, dt: float = 1e-2, mu: float = 0.0):
+                 theta: float = 0.15, dt: float = 1.0, mu: float = 0.0):
         self.sigma = sigma
         self.theta = theta
         self.dt = dt

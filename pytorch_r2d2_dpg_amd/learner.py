@@ -35,7 +35,7 @@ import torch
 from .config import Config
 from .models import make_models
 from .ops.polyak import PolyakUpdater
-from .ops.lstm import lstm_seq_pair, lstm_seq_pair_halfgrad
+from .ops.lstm import _FusedLSTMSeq, lstm_seq_pair, lstm_seq_pair_halfgrad
 from .ops.td import td_targets_and_priorities
 from .parallel.dist import DistContext
 
@@ -318,7 +318,6 @@ class Learner:
         residency instead. Parameter-equivalence is locked by
         test_pair_bwd_step_matches_default."""
         from .ops import hip_ext
-        from .ops.lstm import _FusedLSTMSeq
         cfg = self.cfg
         ext = hip_ext()
         Tn = self.train_len + self.n_step
@@ -469,7 +468,10 @@ class Learner:
         self.opt_critic.zero_grad(set_to_none=zero_to_none)
         for p in cparams:
             g = gmap.get(id(p))
-            assert g is not None, "unmapped critic parameter in pair-bwd path"
+            if g is None:  # a None grad would be silently SKIPPED by the
+                raise RuntimeError(  # all-reduce — fail loudly instead
+                    "pair-bwd path: critic parameter without an assembled "
+                    "gradient (model structure changed?)")
             if p.grad is None:
                 p.grad = g if zero_to_none else g.clone()
             else:

@@ -66,6 +66,12 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
         load_checkpoint(learner, resume)
 
     actors_per_rank = max(1, cfg.actor.num_actors // world)
+    if torch.device(device).type == "cpu":
+        # leave cores for the actor processes: a full-width intra-op thread
+        # pool spin-waits on preempted workers and slows train_step ~80x
+        # (measured 2.45s -> 0.031s/step at 8 cores with 2 actors)
+        torch.set_num_threads(max(2, (os.cpu_count() or 8) // world
+                              - actors_per_rank - 1))
     pool = ActorPool(cfg, actors_per_rank, base_actor_id=rank * actors_per_rank,
                      eval_actors=cfg.actor.eval_actors if rank == 0 else 0)
     pool.start()
@@ -79,11 +85,21 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
     # (same stream) never sees a torn slot (§3.4 single-writer contract:
     # this thread is the only writer).
     stop_ingest = threading.Event()
-    sync_ingest = bool(os.environ.get("R2D2_SYNC_INGEST"))
+    # The threaded ingest is only safe on GPU, where every add_batch op
+    # enqueues on the learner's HIP stream (stream-serialized with its
+    # sampling — §3.4 single-writer contract). On CPU there is no stream:
+    # a background-thread writer races the learner's index_select and the
+    # sum-tree recompute, silently corrupting sampled batches (observed as
+    # an orchestrated CPU pendulum run that never learns while the same
+    # data plane driven synchronously does), so CPU always ingests
+    # synchronously between learner steps.
+    sync_ingest = bool(os.environ.get("R2D2_SYNC_INGEST")) or not replay.is_cuda
 
     def ingest_once():
         pool.drain_episodes()
-        pool.ingest_rings(staging)
+        # bounded: a slow (CPU) learner must not drain the actors' whole
+        # backlog each step — full rings throttle actors via drop-newest
+        pool.ingest_rings(staging, max_per_ring=64)
 
     def ingest_loop():
         if replay.is_cuda:

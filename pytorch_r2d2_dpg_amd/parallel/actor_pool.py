@@ -211,10 +211,11 @@ class ActorPool:
                                     float(prio[i]), obs.shape[1]))
         return out
 
-    def ingest_rings(self, staging) -> int:
+    def ingest_rings(self, staging, max_per_ring: int = 0) -> int:
         """Zero-pickle fast path: memcpy unread ring slots straight into the
-        pinned staging buffers. Called from the learner-side ingest thread."""
-        return staging.ingest_from_rings(self.rings)
+        pinned staging buffers. Called from the learner-side ingest thread
+        (unbounded) or synchronously between CPU learner steps (bounded)."""
+        return staging.ingest_from_rings(self.rings, max_per_ring)
 
     def publish_weights(self, actor_sd: Dict[str, torch.Tensor],
                         critic_sd: Dict[str, torch.Tensor]) -> None:

@@ -58,17 +58,22 @@ class StagingRing:
         return s
 
     @torch.no_grad()
-    def ingest_from_rings(self, rings) -> int:
+    def ingest_from_rings(self, rings, max_per_ring: int = 0) -> int:
         """Copy unread slots from ALL shared-memory SeqRings into pinned
         staging, AGGREGATED into full slots before each dispatch — one
         dispatch is ~15 CUDA enqueues, and thousands of tiny dispatches/s
         from the ingest thread contend with the learner on the HIP runtime
-        lock (plain memcpy with dtype cast; no pickling)."""
+        lock (plain memcpy with dtype cast; no pickling).
+
+        max_per_ring bounds the slots taken per ring per call (0 = all):
+        the synchronous CPU ingest uses it so a slow learner is not starved
+        by draining the actors' whole backlog every step — full rings then
+        throttle the actors through drop-newest backpressure."""
         total = 0
         s = None
         fill = 0
         for ring in rings:
-            ranges = ring.drain_ranges()
+            ranges = ring.drain_ranges(max_per_ring)
             for lo, hi in ranges:
                 off = lo
                 while off < hi:

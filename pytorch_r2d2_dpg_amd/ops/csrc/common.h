@@ -7,6 +7,13 @@
 
 #define DEV_INLINE __device__ __forceinline__
 
+// Host-side post-launch check: a dispatch rejected by the runtime (bad
+// geometry vs launch bounds, LDS overflow, ...) is otherwise SILENT and
+// leaves the output garbage. Returns -3 so callers can fail loudly or
+// fall back.
+#define LSTM_CHECK_LAUNCH() \
+    do { if (hipGetLastError() != hipSuccess) return -3; } while (0)
+
 constexpr int WAVE = 64;
 
 DEV_INLINE float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }

@@ -424,6 +424,7 @@ extern "C" int launch_lstm_fwd_bf16(
             (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh,
             (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
             (__hip_bfloat16*)gates_act, c_seq, T, B, H, stash)));
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -439,6 +440,7 @@ extern "C" int launch_lstm_fwd_f32(
             (const float*)gates_x, (const float*)h0, (const float*)c0,
             (const float*)wh, (float*)h_seq, (float*)c_final,
             (float*)gates_act, c_seq, T, B, H, stash)));
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -458,6 +460,7 @@ extern "C" int launch_lstm_bwd_bf16(
             (const __hip_bfloat16*)c0, (const __hip_bfloat16*)whT,
             (__hip_bfloat16*)dgates, (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0,
             T, B, H, db)));
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -475,5 +478,6 @@ extern "C" int launch_lstm_bwd_f32(
             (const float*)dh_seq, (const float*)dhT, (const float*)dcT,
             (const float*)gates_act, c_seq, (const float*)c0, (const float*)whT,
             (float*)dgates, (float*)dh0, (float*)dc0, T, B, H, db)));
+    LSTM_CHECK_LAUNCH();
     return 0;
 }

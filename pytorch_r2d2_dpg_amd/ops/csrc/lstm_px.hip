@@ -267,6 +267,7 @@ extern "C" int launch_lstm_fwd_px_bf16(
         default:
             return -1;
     }
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -563,6 +564,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
         case 8: lstm_bwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
         default: return -1;
     }
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -607,6 +609,7 @@ extern "C" int launch_lstm_bwd_px2_bf16(
         case 8: lstm_bwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
         default: return -1;
     }
+    LSTM_CHECK_LAUNCH();
     return 0;
 }
 
@@ -831,5 +834,6 @@ extern "C" int launch_lstm_fwd_px2_bf16(
         case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
         default: return -1;
     }
+    LSTM_CHECK_LAUNCH();
     return 0;
 }

@@ -427,9 +427,15 @@ class Learner:
             t_.record_stream(main)
         zero = torch.zeros_like(h_c2)
         whT_c = wh_c.t().contiguous()
-        dg_cf, _, _, dg_q, _, _ = ext.lstm_seq_bwd_pair(
-            dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c,
-            dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
+        try:
+            dg_cf, _, _, dg_q, _, _ = ext.lstm_seq_bwd_pair(
+                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c,
+                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
+        except RuntimeError:  # paired geometry unsupported: two solo K2 calls
+            dg_cf, _, _ = ext.lstm_seq_bwd(
+                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c)
+            dg_q, _, _ = ext.lstm_seq_bwd(
+                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
         self._ev_bw.record(main)
 
         # ---- side: finish the actor chain (dmu -> actor grads), overlapping

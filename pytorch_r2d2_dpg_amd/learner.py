@@ -138,7 +138,9 @@ class Learner:
         # gradient assembly — two concurrent solo bwd grids contend for CU
         # residency instead (+3% measured; parameter-equivalent to the
         # autograd path, test_pair_bwd_step_matches_default).
+        _H = cfg.model.hidden_size
         self._pair_bwd = (overlap_ok and self.dtype == torch.bfloat16
+                          and _H >= 64 and (_H & (_H - 1)) == 0  # px geometry
                           and not _os.environ.get("R2D2_NO_PAIR_BWD"))
         if self._pair_bwd:
             self._ev_bw = torch.cuda.Event()

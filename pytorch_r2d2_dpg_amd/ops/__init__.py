@@ -15,7 +15,6 @@ eager PyTorch (the round-end harness records which .so files actually loaded).
 
 from __future__ import annotations
 
-import os
 
 import torch
 

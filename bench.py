@@ -38,9 +38,16 @@ from pytorch_r2d2_dpg_amd.replay.gpu_buffer import GpuSequenceReplay
 
 def flagship_config(args) -> Config:
     cfg = Config()
-    cfg.env.name = "humanoid-run"
-    cfg.model.hidden_size = args.hidden
-    cfg.model.trunk_sizes = (args.hidden, args.hidden)
+    cfg.env.name = args.env
+    if args.env == "quadruped-run-pixels":
+        # BASELINE.json config 4 (configs/quadruped_pixel.toml): Conv+LSTM
+        # encoder, uint8 pixel obs resident in the HBM shard
+        cfg.env.pixel = True
+        cfg.model.hidden_size = args.hidden
+        cfg.model.trunk_sizes = (args.hidden,)
+    else:
+        cfg.model.hidden_size = args.hidden
+        cfg.model.trunk_sizes = (args.hidden, args.hidden)
     cfg.model.dtype = args.dtype
     cfg.replay.seq_len = args.seq_len
     cfg.replay.burn_in = args.burn_in
@@ -62,7 +69,11 @@ def fill_replay_synthetic(replay: GpuSequenceReplay, spec, cfg: Config, n_seqs: 
     done = 0
     while done < n_seqs:
         n = min(chunk, n_seqs - done)
-        obs = torch.randn(n, L, *spec.obs_shape, generator=g)
+        if spec.pixel:
+            obs = torch.randint(0, 256, (n, L) + tuple(spec.obs_shape),
+                                dtype=torch.uint8, generator=g)
+        else:
+            obs = torch.randn(n, L, *spec.obs_shape, generator=g)
         act = torch.randn(n, L, spec.action_dim, generator=g).clamp_(-1, 1)
         rew = torch.rand(n, L, generator=g)
         dn = torch.zeros(n, L)
@@ -75,14 +86,24 @@ def fill_replay_synthetic(replay: GpuSequenceReplay, spec, cfg: Config, n_seqs: 
 def main() -> int:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=256)
+    # defaults give a multi-second timed region (~2.5 s at 5.8 ms/step) so
+    # run-to-run variance is observable, while still finishing in well under
+    # a minute including warmup/fill
+    p.add_argument("--steps", type=int, default=400)
+    p.add_argument("--warmup", type=int, default=40)
+    p.add_argument("--env", type=str, default="humanoid-run",
+                   choices=["humanoid-run", "cheetah-run", "quadruped-run-pixels"],
+                   help="humanoid-run = flagship (BASELINE config 3); "
+                        "quadruped-run-pixels = Conv+LSTM pixel config 4")
+    p.add_argument("--batch", type=int, default=None,
+                   help="sequences per rank per step (default: 256; pixel: 32 "
+                        "to match configs/quadruped_pixel.toml)")
     p.add_argument("--seq-len", type=int, default=80)
     p.add_argument("--burn-in", type=int, default=40)
-    p.add_argument("--hidden", type=int, default=512)
+    p.add_argument("--hidden", type=int, default=None,
+                   help="LSTM hidden (default: 512; pixel: 256)")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
-    p.add_argument("--replay-capacity", type=int, default=8192)
+    p.add_argument("--replay-capacity", type=int, default=None)
     p.add_argument("--fp8-obs", action="store_true",
                    help="store observations fp8 e4m3 in the replay shard (K7)")
     p.add_argument("--device", type=str, default="")
@@ -91,6 +112,13 @@ def main() -> int:
                         "capture serializes the multi-stream overlap)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
     args = p.parse_args()
+    pixel = args.env == "quadruped-run-pixels"
+    if args.batch is None:
+        args.batch = 32 if pixel else 256
+    if args.hidden is None:
+        args.hidden = 256 if pixel else 512
+    if args.replay_capacity is None:
+        args.replay_capacity = 2048 if pixel else 8192
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     dist = DistContext.from_env()
@@ -113,6 +141,7 @@ def main() -> int:
         obs_shape=spec.obs_shape, action_dim=spec.action_dim,
         hidden_size=cfg.model.hidden_size, per_alpha=cfg.replay.per_alpha,
         device=device, seed=cfg.learner.seed + 17 * rank,
+        obs_dtype=torch.uint8 if spec.pixel else torch.float32,
         store_dtype=torch.bfloat16 if args.dtype == "bf16" else torch.float32,
         fp8_obs=args.fp8_obs,
     )
@@ -166,9 +195,16 @@ def main() -> int:
     n_gpus = world if world > 1 else args.gpus
     value = n_gpus * B * T / (elapsed / args.steps)
 
+    if pixel:
+        model_desc = ("R2D2-DPG quadruped-run-pixels (84x84x3 uint8, act 12), "
+                      "Conv(32,64,64)+LSTM H=" + str(args.hidden))
+    else:
+        model_desc = ("R2D2-DPG " + args.env + " (obs "
+                      + str(spec.obs_shape[0]) + ", act " + str(spec.action_dim)
+                      + "), LSTM H=" + str(args.hidden) + " trunk 2x" + str(args.hidden))
     if rank == 0:
         out = {
-            "metric": "learner transitions/sec (whole node), humanoid-run",
+            "metric": "learner transitions/sec (whole node), " + args.env,
             "value": round(value, 1),
             "unit": "transitions/s",
             "n_gpus": n_gpus,
@@ -181,8 +217,7 @@ def main() -> int:
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "model": "R2D2-DPG humanoid-run (obs 67, act 21), LSTM H="
-                         + str(args.hidden) + " trunk 2x" + str(args.hidden),
+                "model": model_desc,
                 "global_batch": B * n_gpus,
                 "seq_len": args.seq_len,
                 "burn_in": args.burn_in,

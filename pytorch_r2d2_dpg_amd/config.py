@@ -6,8 +6,8 @@ this framework ships the five BASELINE.json configs under ``configs/``).
 
 Numeric defaults follow SURVEY.md §2.1 (R2D2 paper values with DPG-style soft
 target updates): seq 80 / burn-in 40 / overlap 40, n-step 5, γ 0.997,
-priority p = η·max|δ| + (1−η)·mean|δ| with η = 0.9, PER α = 0.9, β = 0 (no IS
-correction by default), Polyak τ = 5e-3, Adam lr 1e-4.
+priority p = η·max|δ| + (1−η)·mean|δ| with η = 0.9, PER α = 0.9, β = 0.6
+(the paper pairs α = 0.9 with IS exponent 0.6), Polyak τ = 5e-3, Adam lr 1e-4.
 """
 
 from __future__ import annotations
@@ -49,7 +49,9 @@ class ReplayConfig:
     gamma: float = 0.997
     priority_eta: float = 0.9       # p = eta*max|d| + (1-eta)*mean|d|
     per_alpha: float = 0.9
-    per_beta: float = 0.0           # importance-sampling exponent (R2D2 used 0)
+    per_beta: float = 0.6           # importance-sampling exponent. The R2D2 paper
+                                    # ran alpha=0.9 WITH beta=0.6 (Ape-X values);
+                                    # set 0 to disable IS correction entirely.
     min_fill: int = 256             # sequences before learning starts
     reward_scale: float = 1.0       # rewards are scaled at actor-push time
     fp8_obs: bool = False           # store observations as fp8 e4m3 in the GPU shard

@@ -6,10 +6,16 @@ target-net unroll → fused n-step TD targets (K3) → critic update → actor
 update (ascend grad-Q through the critic) → fused Polyak (K4) → priority
 write-back (K6) → periodic weight publish / checkpoint.
 
+Padded tail steps of episode-flushed sequences (``batch["length"]`` <
+stored length) are masked out of the critic loss, the actor loss and the
+K3 priority reduction; time-limit truncation bootstraps through the
+repeated final observation (replay/sequence.py).
+
 Data-parallel: each rank runs this identically on its own shard; gradients
 are all-reduced as one fused bucket over RCCL/xGMI after each backward
 (parallel/dist.py) before the Adam step — DP ranks therefore hold identical
-weights at all times; rank 0 alone checkpoints/logs.
+weights at all times; rank 0 alone checkpoints/logs. Stream-ordering
+contracts for the multi-stream execution are documented in docs/STREAMS.md.
 
 Execution (GPU): independent unrolls share paired (grid.z=2) column-split
 launches — the two burn-ins, the target-actor unroll with the online
@@ -20,16 +26,21 @@ two gradient all-reduces are issued on the main stream in a fixed order.
 ``capture_graph()`` can record the whole update into a hipGraph, but eager
 + stream overlap measures faster at these sizes (steady-state device
 occupancy is already ~92%), so it is not the default.
+
+The step is ONE code path (`_step_core`): shared unpack → shared burn-in →
+a pluggable middle (`_mid_pair_bwd` = manual paired-K2 backward, default on
+GPU; `_mid_autograd` = plain autograd, CPU + fallback) → shared tail
+(all-reduce, clip, optimizer steps, Polyak, priority write-back). The two
+middles are parameter-equivalent (test_pair_bwd_step_matches_default).
 """
 
 from __future__ import annotations
 
 import contextlib
-import time
+import os
 from dataclasses import dataclass
-from typing import Dict, Optional
+from typing import Dict, Optional, Tuple
 
-import numpy as np
 import torch
 
 from .config import Config
@@ -110,7 +121,6 @@ class Learner:
         # stream-overlap machinery (GPU only): the actor chain runs on a side
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
-        import os as _os
         # Stream overlap pays at B <= 256 (glue hides behind the LSTM
         # chains, +20% measured); at larger batches two concurrent
         # column-split grids contend for CU residency and crawl, so run the
@@ -124,7 +134,7 @@ class Learner:
         overlap_ok = (self.device.type == "cuda"
                       and cfg.learner.batch_size <= 256
                       and self.train_len <= 48
-                      and not _os.environ.get("R2D2_NO_OVERLAP"))
+                      and not os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()
             self._ev_ba = torch.cuda.Event()   # burn-in done (main -> side)
@@ -141,14 +151,22 @@ class Learner:
         _H = cfg.model.hidden_size
         self._pair_bwd = (overlap_ok and self.dtype == torch.bfloat16
                           and _H >= 64 and (_H & (_H - 1)) == 0  # px geometry
-                          and not _os.environ.get("R2D2_NO_PAIR_BWD"))
+                          and not os.environ.get("R2D2_NO_PAIR_BWD"))
         if self._pair_bwd:
             self._ev_bw = torch.cuda.Event()
             self._ev_ab2 = torch.cuda.Event()
 
+        # Column-split spin-timeout surveillance: a px kernel that
+        # times out its bounded spin flags a device error word and PROCEEDS
+        # with stale data — poll the word every N steps (one tiny D2H read)
+        # and FAIL LOUDLY instead of training on corrupted numerics.
+        # R2D2_PX_ERR_WARN=1 downgrades to a warning + counter.
+        self._px_check_every = int(os.environ.get("R2D2_PX_CHECK_EVERY", "50"))
+        self.px_err_total = 0
+
         # optional hipEvent phase timers (SURVEY §5 tracing): R2D2_PHASE_TIMERS=1
         self._phase_events = None
-        if self.device.type == "cuda" and _os.environ.get("R2D2_PHASE_TIMERS"):
+        if self.device.type == "cuda" and os.environ.get("R2D2_PHASE_TIMERS"):
             self._phase_events = {k: (torch.cuda.Event(enable_timing=True),
                                       torch.cuda.Event(enable_timing=True))
                                   for k in ("burn_in", "target", "critic", "actor_opt")}
@@ -164,6 +182,72 @@ class Learner:
         if self._phase_events:
             self._phase_events[key][which].record()
 
+    # --------------------------------------------------------- shared pieces
+    def _unpack(self, batch: Dict[str, torch.Tensor]):
+        """Batch unpack + pad-mask construction, shared by both middles.
+
+        Returns (obs [L,B,...], action [L,B,A], reward/done [B,L], weight [B],
+        h_a, c_a, h_c, c_c [B,H], vmask [B,T] fp32 or None, valid [B] i32 or
+        None). vmask/valid are None when the batch has no padded sequences
+        (bench fast path — replay returns length=None then)."""
+        obs = batch["obs"].transpose(0, 1)                         # [L, B, ...]
+        if obs.dtype != torch.uint8:  # pixel obs stay uint8 (trunk normalizes)
+            obs = obs.to(self.dtype)
+        action = batch["action"].transpose(0, 1).to(self.dtype)    # [L, B, A]
+        reward = batch["reward"]                                   # [B, L] fp32
+        done = batch["done"]                                       # [B, L] fp32
+        weight = batch["weight"]                                   # [B] fp32
+        st = batch["state"].to(self.dtype)                         # [B, 4, H]
+        h_a, c_a = st[:, 0].contiguous(), st[:, 1].contiguous()
+        h_c, c_c = st[:, 2].contiguous(), st[:, 3].contiguous()
+        length = batch.get("length")
+        if length is not None:
+            T = self.train_len
+            valid = (length.to(torch.int32) - self.burn_in).clamp(1, T)
+            vmask = (torch.arange(T, device=obs.device).unsqueeze(0)
+                     < valid.to(torch.int64).unsqueeze(1)).to(torch.float32)
+        else:
+            valid = vmask = None
+        return obs, action, reward, done, weight, h_a, c_a, h_c, c_c, vmask, valid
+
+    def _burn_in(self, obs, action, h_a, c_a, h_c, c_c):
+        """BOTH no-grad burn-in unrolls in ONE paired column-split launch
+        (grid.z = 2) — a single grid co-dispatches uniformly, so the two
+        halves genuinely parallelize (two px launches on separate streams
+        contend for CU residency instead)."""
+        bi = self.burn_in
+        if bi == 0:
+            return h_a, c_a, h_c, c_c
+        with torch.no_grad():
+            feat_a = self.actor.trunk(obs[:bi])
+            feat_c = torch.cat([self.critic.trunk(obs[:bi]), action[:bi]], dim=-1)
+            la, lc = self.actor.lstm, self.critic.lstm
+            _, h_a2, c_a2, _, h_c2, c_c2 = lstm_seq_pair(
+                feat_a, h_a, c_a, la.weight_x, la.weight_h, la.bias,
+                feat_c, h_c, c_c, lc.weight_x, lc.weight_h, lc.bias)
+        return h_a2, c_a2, h_c2, c_c2
+
+    def _critic_loss(self, y, q_online_bt, weight, vmask):
+        """Weighted MSE over REAL steps only (pad steps carry no loss)."""
+        if vmask is None:
+            return (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
+        return ((weight.unsqueeze(1) * vmask * (y - q_online_bt) ** 2).sum()
+                / vmask.sum())
+
+    def _actor_loss(self, q_pi, vmask):
+        """-mean Q over REAL steps (q_pi is [T, B]; vmask is [B, T])."""
+        if vmask is None:
+            return -q_pi.float().mean()
+        m_tb = vmask.t()
+        return -(q_pi.float() * m_tb).sum() / m_tb.sum()
+
+    def _bump_weight_version(self):
+        """Invalidate the HIP extension's fragment-major weight-pack cache
+        after the optimizer/Polyak writes (ops/csrc/bindings.cpp)."""
+        if self.device.type == "cuda":
+            from .ops import hip_ext
+            hip_ext().bump_weight_version()
+
     # ---------------------------------------------------------- core update
     def _step_core(self, batch: Dict[str, torch.Tensor],
                    update_priorities: bool = True,
@@ -177,60 +261,94 @@ class Learner:
         unrolls is nearly free parallelism). The CPU path runs the identical
         math on one stream.
         """
-        if getattr(self, "_pair_bwd", False):
-            return self._step_core_pair_bwd(batch, update_priorities, zero_to_none)
         cfg = self.cfg
-        Tn = self.train_len + self.n_step
-        bi = self.burn_in
-        T = self.train_len
+        (obs, action, reward, done, weight,
+         h_a, c_a, h_c, c_c, vmask, valid) = self._unpack(batch)
         overlap = self.device.type == "cuda" and self._side_stream is not None
-
-        obs = batch["obs"].transpose(0, 1)                         # [L, B, ...]
-        if obs.dtype != torch.uint8:  # pixel obs stay uint8 (trunk normalizes)
-            obs = obs.to(self.dtype)
-        action = batch["action"].transpose(0, 1).to(self.dtype)    # [L, B, A]
-        reward = batch["reward"]                                   # [B, L] fp32
-        done = batch["done"]                                       # [B, L] fp32
-        weight = batch["weight"]                                   # [B] fp32
-        st = batch["state"].to(self.dtype)                         # [B, 4, H]
-        h_a, c_a = st[:, 0].contiguous(), st[:, 1].contiguous()
-        h_c, c_c = st[:, 2].contiguous(), st[:, 3].contiguous()
-
-        s1 = self._side_stream
-        ev_ba = self._ev_ba
-        ev_ab = self._ev_ab
-
         main = torch.cuda.current_stream() if overlap else None
+        s1 = self._side_stream
         if overlap:
             s1.wait_stream(main)
 
         self._pe("burn_in", 0)
-        # ---- burn-in: BOTH no-grad unrolls in ONE paired column-split
-        # launch (grid.z = 2) — a single grid co-dispatches uniformly, so
-        # the two halves genuinely parallelize (two px launches on separate
-        # streams contend for CU residency instead)
-        if bi > 0:
-            with torch.no_grad():
-                feat_a = self.actor.trunk(obs[:bi])
-                feat_c = torch.cat([self.critic.trunk(obs[:bi]), action[:bi]], dim=-1)
-                la, lc = self.actor.lstm, self.critic.lstm
-                _, h_a2, c_a2, _, h_c2, c_c2 = lstm_seq_pair(
-                    feat_a, h_a, c_a, la.weight_x, la.weight_h, la.bias,
-                    feat_c, h_c, c_c, lc.weight_x, lc.weight_h, lc.bias)
-        else:
-            h_a2, c_a2, h_c2, c_c2 = h_a, c_a, h_c, c_c
+        h_a2, c_a2, h_c2, c_c2 = self._burn_in(obs, action, h_a, c_a, h_c, c_c)
         if overlap:
-            ev_ba.record(main)
+            self._ev_ba.record(main)
+
+        mid = self._mid_pair_bwd if getattr(self, "_pair_bwd", False) \
+            else self._mid_autograd
+        critic_loss, actor_loss, actor_grads, prio, q_online_bt, join_ev = mid(
+            obs, action, reward, done, weight, h_a2, c_a2, h_c2, c_c2,
+            vmask, valid, overlap, main, s1, zero_to_none)
+
+        # ---- shared tail: all-reduce -> clip -> join -> optimizer steps ----
+        cparams = list(self.critic.parameters())
+        self.dist.allreduce_grads(cparams, key="critic")
+        if cfg.learner.max_grad_norm > 0:
+            torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
+        # the side stream reads critic params: join before the optimizer writes
+        if join_ev is not None:
+            main.wait_event(join_ev)
+        self.opt_critic.step()
+
+        self._pe("critic", 1)
+        self._pe("actor_opt", 0)
+        # ---- actor optimizer (grads computed on the side stream)
+        for p, g in zip(self._actor_params, actor_grads):
+            if overlap:
+                # g was allocated on the side stream but is consumed by the
+                # all-reduce/optimizer on main: mark the cross-stream use so
+                # the caching allocator does not recycle g for a later
+                # side-stream allocation while main's reads are in flight
+                g.record_stream(main)
+            if p.grad is None:
+                p.grad = g.clone() if not zero_to_none else g
+            else:
+                p.grad.copy_(g)
+        self.dist.allreduce_grads(self._actor_params, key="actor")
+        if cfg.learner.max_grad_norm > 0:
+            torch.nn.utils.clip_grad_norm_(self._actor_params, cfg.learner.max_grad_norm)
+        self.opt_actor.step()
+
+        # ---- Polyak soft target update (K4: one fused launch per net on GPU)
+        # (branch is Python-side: under graph capture target_update_every == 1)
+        if self.step_count % cfg.learner.target_update_every == 0:
+            self.polyak_actor(cfg.learner.tau)
+            self.polyak_critic(cfg.learner.tau)
+        self._bump_weight_version()
+
+        self._pe("actor_opt", 1)
+        # ---- priority write-back (stays on-GPU with the K6 tree kernel)
+        if update_priorities and self.replay is not None and batch.get("idx") is not None:
+            self.replay.update_priorities(batch["idx"], prio)
+
+        return {
+            "critic_loss": critic_loss.detach(),
+            "actor_loss": actor_loss.detach(),
+            "mean_q": q_online_bt.detach().mean(),
+            "mean_priority": prio.mean(),
+        }
+
+    # -------------------------------------------- middle: autograd backward
+    def _mid_autograd(self, obs, action, reward, done, weight,
+                      h_a2, c_a2, h_c2, c_c2, vmask, valid,
+                      overlap, main, s1, zero_to_none):
+        """Plain-autograd middle (CPU path + GPU fallback): actor chain on
+        the side stream via autograd.grad; critic loss via .backward()."""
+        cfg = self.cfg
+        bi, T = self.burn_in, self.train_len
+        Tn = T + self.n_step
+
         with torch.cuda.stream(s1) if overlap else _null_ctx():
             if overlap:
-                s1.wait_event(ev_ba)
+                s1.wait_event(self._ev_ba)
             # actor chain through the PRE-update critic
             mu, _ = self.actor(obs[bi:bi + T], (h_a2.detach(), c_a2.detach()))
             q_pi, _ = self.critic(obs[bi:bi + T], mu, (h_c2.detach(), c_c2.detach()))
-            actor_loss = -q_pi.float().mean()
+            actor_loss = self._actor_loss(q_pi, vmask)
             actor_grads = torch.autograd.grad(actor_loss, self._actor_params)
             if overlap:
-                ev_ab.record(s1)
+                self._ev_ab.record(s1)
 
         self._pe("burn_in", 1)
         self._pe("target", 0)
@@ -262,108 +380,48 @@ class Learner:
             y, td_abs, prio = td_targets_and_priorities(
                 reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),
                 self.gamma, self.n_step, self.eta,
-                rescale=cfg.learner.value_rescale,
+                rescale=cfg.learner.value_rescale, valid_n=valid,
             )
         self._pe("target", 1)
         self._pe("critic", 0)
-        critic_loss = (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
+        critic_loss = self._critic_loss(y, q_online_bt, weight, vmask)
         self.opt_critic.zero_grad(set_to_none=zero_to_none)
         critic_loss.backward()
-        self.dist.allreduce_grads(self.critic.parameters(), key="critic")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(self.critic.parameters(), cfg.learner.max_grad_norm)
-        # the actor chain reads critic params: join before the optimizer writes
-        if overlap:
-            main.wait_event(ev_ab)
-        self.opt_critic.step()
+        return (critic_loss, actor_loss, actor_grads, prio, q_online_bt,
+                self._ev_ab if overlap else None)
 
-        self._pe("critic", 1)
-        self._pe("actor_opt", 0)
-        # ---- actor optimizer (grads computed on the side stream)
-        for p, g in zip(self._actor_params, actor_grads):
-            if p.grad is None:
-                p.grad = g.clone() if not zero_to_none else g
-            else:
-                p.grad.copy_(g)
-        self.dist.allreduce_grads(self._actor_params, key="actor")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(self._actor_params, cfg.learner.max_grad_norm)
-        self.opt_actor.step()
-
-        # ---- Polyak soft target update (K4: one fused launch per net on GPU)
-        # (branch is Python-side: under graph capture target_update_every == 1)
-        if self.step_count % cfg.learner.target_update_every == 0:
-            self.polyak_actor(cfg.learner.tau)
-            self.polyak_critic(cfg.learner.tau)
-
-        self._pe("actor_opt", 1)
-        # ---- priority write-back (stays on-GPU with the K6 tree kernel)
-        if update_priorities and self.replay is not None and batch.get("idx") is not None:
-            self.replay.update_priorities(batch["idx"], prio)
-
-        return {
-            "critic_loss": critic_loss.detach(),
-            "actor_loss": actor_loss.detach(),
-            "mean_q": q_online_bt.detach().mean(),
-            "mean_priority": prio.mean(),
-        }
-
-    # ------------------------------------------------- paired-backward step
-    def _step_core_pair_bwd(self, batch: Dict[str, torch.Tensor],
-                            update_priorities: bool = True,
-                            zero_to_none: bool = True) -> Dict[str, torch.Tensor]:
-        """Default GPU step at bf16 + overlap (R2D2_NO_PAIR_BWD=1 falls back
-        to _step_core): identical math, but the critic-loss backward and the
-        actor-chain critic backward run as ONE paired (grid.z=2) K2 launch
-        with the surrounding dense gradients (gates GEMMs, trunk, heads)
-        assembled manually — two concurrent solo bwd grids contend for CU
-        residency instead. Parameter-equivalence is locked by
+    # ------------------------------------------ middle: paired-K2 backward
+    def _mid_pair_bwd(self, obs, action, reward, done, weight,
+                      h_a2, c_a2, h_c2, c_c2, vmask, valid,
+                      overlap, main, s1, zero_to_none):
+        """Default GPU middle at bf16 + overlap (R2D2_NO_PAIR_BWD=1 falls
+        back to autograd): identical math, but the critic-loss backward and
+        the actor-chain critic backward run as ONE paired (grid.z=2) K2
+        launch with the surrounding dense gradients (gates GEMMs, trunk,
+        heads) assembled manually — two concurrent solo bwd grids contend
+        for CU residency instead. Parameter-equivalence is locked by
         test_pair_bwd_step_matches_default."""
         from .ops import hip_ext
         cfg = self.cfg
         ext = hip_ext()
-        Tn = self.train_len + self.n_step
-        bi = self.burn_in
-        T = self.train_len
+        bi, T = self.burn_in, self.train_len
+        Tn = T + self.n_step
         H = self.actor.hidden_size
-
-        obs = batch["obs"].transpose(0, 1)
-        if obs.dtype != torch.uint8:
-            obs = obs.to(self.dtype)
-        action = batch["action"].transpose(0, 1).to(self.dtype)
-        reward = batch["reward"]
-        done = batch["done"]
-        weight = batch["weight"]
-        st = batch["state"].to(self.dtype)
-        h_a, c_a = st[:, 0].contiguous(), st[:, 1].contiguous()
-        h_c, c_c = st[:, 2].contiguous(), st[:, 3].contiguous()
-
-        main = torch.cuda.current_stream()
-        s1 = self._side_stream
-        s1.wait_stream(main)
-
-        self._pe("burn_in", 0)
-        if bi > 0:
-            with torch.no_grad():
-                feat_ab = self.actor.trunk(obs[:bi])
-                feat_cb = torch.cat([self.critic.trunk(obs[:bi]), action[:bi]], dim=-1)
-                la, lcb = self.actor.lstm, self.critic.lstm
-                _, h_a2, c_a2, _, h_c2, c_c2 = lstm_seq_pair(
-                    feat_ab, h_a, c_a, la.weight_x, la.weight_h, la.bias,
-                    feat_cb, h_c, c_c, lcb.weight_x, lcb.weight_h, lcb.bias)
-        else:
-            h_a2, c_a2, h_c2, c_c2 = h_a, c_a, h_c, c_c
-        self._ev_ba.record(main)
 
         lc = self.critic.lstm
         wx_c, wh_c, b_c = lc.weight_x, lc.weight_h, lc.bias
         obs_w = obs[bi:bi + T]
         c_c2d = c_c2.detach().contiguous()
+        w_head = self.critic.head.weight          # [1, H]
+        b_head = self.critic.head.bias
 
-        # ---- side stream: actor chain FORWARD only + the q_pi head seed.
+        # ---- side stream: actor chain FORWARD only + the q_pi loss seed.
         # The critic unroll here is a RAW stash forward (its weight grads
         # are discarded by design — pre-update-critic DDPG), so only dmu is
-        # needed from its backward, which the paired K2 below provides.
+        # needed from its backward, which the paired K2 below provides. The
+        # loss seed d(actor_loss)/d(hs_q) is ANALYTIC (-m/Σm ⊗ w_head): no
+        # autograd graph touches the critic head here, which keeps the head
+        # params' AccumulateGrad nodes on the main stream (docs/STREAMS.md).
         with torch.cuda.stream(s1):
             s1.wait_event(self._ev_ba)
             la = self.actor.lstm
@@ -375,14 +433,20 @@ class Learner:
             with torch.no_grad():
                 feat_qc = self.critic.trunk(obs_w)
                 x_q = torch.cat([feat_qc, mu.detach()], dim=-1)
-                gx_q = torch.addmm(b_c, x_q.reshape(T * x_q.shape[1], -1),
+                B_ = x_q.shape[1]
+                gx_q = torch.addmm(b_c, x_q.reshape(T * B_, -1),
                                    wx_c.t()).view(T, -1, 4 * H).contiguous()
                 hs_q, _, ga_q, cs_q = ext.lstm_seq_fwd(
                     gx_q, h_c2.detach().contiguous(), c_c2d, wh_c.contiguous())
-            hs_q_leaf = hs_q.detach().requires_grad_(True)
-            q_pi = self.critic.head(hs_q_leaf.reshape(-1, H)).view(T, -1)
-            actor_loss = -q_pi.float().mean()
-            dhs_q, = torch.autograd.grad(actor_loss, [hs_q_leaf])
+                q_pi = (hs_q.reshape(-1, H) @ w_head.t() + b_head).view(T, B_)
+                actor_loss = self._actor_loss(q_pi, vmask)
+                if vmask is None:
+                    dhs_q = (w_head.mul(-1.0 / (T * B_)).to(hs_q.dtype)
+                             .view(1, 1, H).expand(T, B_, H).contiguous())
+                else:
+                    m_tb = vmask.t()
+                    dhs_q = ((m_tb / -m_tb.sum()).unsqueeze(-1).to(hs_q.dtype)
+                             * w_head.view(1, 1, H)).contiguous()
             self._ev_ab.record(s1)
 
         self._pe("burn_in", 1)
@@ -414,11 +478,11 @@ class Learner:
             y, td_abs, prio = td_targets_and_priorities(
                 reward[:, bi:], done[:, bi:], q_boot, q_online_bt.detach(),
                 self.gamma, self.n_step, self.eta,
-                rescale=cfg.learner.value_rescale,
+                rescale=cfg.learner.value_rescale, valid_n=valid,
             )
         self._pe("target", 1)
         self._pe("critic", 0)
-        critic_loss = (weight.unsqueeze(1) * (y - q_online_bt) ** 2).mean()
+        critic_loss = self._critic_loss(y, q_online_bt, weight, vmask)
         head_p = [self.critic.head.weight, self.critic.head.bias]
         dhs_cf, dw_head, db_head = torch.autograd.grad(
             critic_loss, [hs_cf_leaf] + head_p)
@@ -428,16 +492,15 @@ class Learner:
         for t_ in (dhs_q, ga_q, cs_q):
             t_.record_stream(main)
         zero = torch.zeros_like(h_c2)
-        whT_c = wh_c.t().contiguous()
         try:
             dg_cf, _, _, dg_q, _, _ = ext.lstm_seq_bwd_pair(
-                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c,
-                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
+                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, wh_c,
+                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, wh_c)
         except RuntimeError:  # paired geometry unsupported: two solo K2 calls
             dg_cf, _, _ = ext.lstm_seq_bwd(
-                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, whT_c)
+                dhs_cf.contiguous(), zero, zero, ga_cf, cs_cf, c_c2d, wh_c)
             dg_q, _, _ = ext.lstm_seq_bwd(
-                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, whT_c)
+                dhs_q.contiguous(), zero, zero, ga_q, cs_q, c_c2d, wh_c)
         self._ev_bw.record(main)
 
         # ---- side: finish the actor chain (dmu -> actor grads), overlapping
@@ -472,9 +535,8 @@ class Learner:
         gmap[id(b_c)] = db_c
         gmap[id(head_p[0])] = dw_head
         gmap[id(head_p[1])] = db_head
-        cparams = list(self.critic.parameters())
         self.opt_critic.zero_grad(set_to_none=zero_to_none)
-        for p in cparams:
+        for p in self.critic.parameters():
             g = gmap.get(id(p))
             if g is None:  # a None grad would be silently SKIPPED by the
                 raise RuntimeError(  # all-reduce — fail loudly instead
@@ -484,40 +546,8 @@ class Learner:
                 p.grad = g if zero_to_none else g.clone()
             else:
                 p.grad.copy_(g)
-        self.dist.allreduce_grads(cparams, key="critic")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
-        # the side stream reads critic params (head fwd, wx_c): join before write
-        main.wait_event(self._ev_ab2)
-        self.opt_critic.step()
-
-        self._pe("critic", 1)
-        self._pe("actor_opt", 0)
-        for p, g in zip(self._actor_params, actor_grads):
-            if p.grad is None:
-                p.grad = g.clone() if not zero_to_none else g
-            else:
-                p.grad.copy_(g)
-        self.dist.allreduce_grads(self._actor_params, key="actor")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(self._actor_params,
-                                           cfg.learner.max_grad_norm)
-        self.opt_actor.step()
-
-        if self.step_count % cfg.learner.target_update_every == 0:
-            self.polyak_actor(cfg.learner.tau)
-            self.polyak_critic(cfg.learner.tau)
-
-        self._pe("actor_opt", 1)
-        if update_priorities and self.replay is not None and batch.get("idx") is not None:
-            self.replay.update_priorities(batch["idx"], prio)
-
-        return {
-            "critic_loss": critic_loss.detach(),
-            "actor_loss": actor_loss.detach(),
-            "mean_q": q_online_bt.detach().mean(),
-            "mean_priority": prio.mean(),
-        }
+        return (critic_loss, actor_loss, actor_grads, prio, q_online_bt,
+                self._ev_ab2)
 
     # ------------------------------------------------------------ graph cap
     def capture_graph(self, example_batch: Dict[str, torch.Tensor]) -> None:
@@ -561,6 +591,7 @@ class Learner:
         else:
             stats = self._step_core(batch)
         self.step_count += 1
+        self._check_px_errors()
         B = batch["obs"].shape[0]
         if self.collect_stats:
             return StepStats(
@@ -571,6 +602,28 @@ class Learner:
                 transitions=B * self.train_len,
             )
         return StepStats(0.0, 0.0, 0.0, 0.0, B * self.train_len)
+
+    def _check_px_errors(self):
+        """Poll the column-split spin-timeout error word every
+        ``_px_check_every`` steps and fail loudly on corruption (the kernels
+        proceed with stale data after a timed-out spin; silently training on
+        that is the failure mode ADVICE r1 flagged)."""
+        if (self.device.type != "cuda" or not self._px_check_every
+                or self.step_count % self._px_check_every != 0):
+            return
+        from .ops import hip_ext
+        err = int(hip_ext().lstm_px_error())
+        if err:
+            self.px_err_total += err
+            msg = (f"column-split LSTM spin timeout flagged (code {err}) within "
+                   f"the last {self._px_check_every} learner steps — numerics "
+                   "of the affected steps are unreliable")
+            if os.environ.get("R2D2_PX_ERR_WARN"):
+                import warnings
+                warnings.warn(msg)
+            else:
+                raise RuntimeError(
+                    msg + " (set R2D2_PX_ERR_WARN=1 to continue anyway)")
 
     # -------------------------------------------------------------- weights
     def actor_state_for_sync(self) -> Dict[str, torch.Tensor]:

@@ -25,8 +25,8 @@ extern "C" int launch_lstm_bwd_f32(const void*, const void*, const void*, const 
                                    const float*, const void*, const void*, void*, void*,
                                    void*, int, int, int, hipStream_t);
 extern "C" void launch_td_targets(const float*, const float*, const float*, const float*,
-                                  float*, float*, float*, int, int, int, int, float, float,
-                                  int, float, hipStream_t);
+                                  const int*, float*, float*, float*, int, int, int, int,
+                                  float, float, int, float, hipStream_t);
 extern "C" void launch_polyak_multi(const unsigned long long*, const long*, int, long,
                                     float, int, hipStream_t);
 extern "C" void launch_sumtree_sample(const float*, int, int, unsigned long long,
@@ -128,6 +128,55 @@ torch::Tensor pack_fragment_major(const torch::Tensor& w) {
         .contiguous();
 }
 
+// ---- packed-weight cache ---------------------------------------------------
+// The fragment-major packs are pure functions of the weight VALUES, which
+// change only at optimizer/Polyak writes; the learner bumps the version once
+// per step after those writes (learner.py::_bump_weight_version) and every
+// LSTM entry point reuses the cached pack until then — the flagship step
+// packs the critic Wh 4x and its transpose 1-2x otherwise.
+// Stream-ordering contract (docs/STREAMS.md): a cache MISS enqueues the
+// repack on the caller's current stream; consumers on other streams are
+// already ordered behind it by the learner's burn-in -> ev_ba fencing (the
+// first pack of each step happens on the main stream before ev_ba). Entries
+// are held by the map, so the caching allocator never recycles them.
+// Keyed by data_ptr: weights are long-lived parameters whose storage is
+// stable; a version bump also clears entries whose shape changed.
+struct PackEntry { uint64_t version; torch::Tensor packed; };
+uint64_t g_weight_version = 0;
+std::unordered_map<const void*, PackEntry> g_pack_cache;    // w -> pack(w)
+std::unordered_map<const void*, PackEntry> g_pack_cache_T;  // w -> pack(w^T)
+
+void bump_weight_version() {
+    ++g_weight_version;
+    if (g_pack_cache.size() > 512) g_pack_cache.clear();      // leak guard
+    if (g_pack_cache_T.size() > 512) g_pack_cache_T.clear();
+}
+
+bool stream_capturing() {
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    (void)hipStreamIsCapturing(cur_stream(), &st);
+    return st != hipStreamCaptureStatusNone;
+}
+
+torch::Tensor pack_cached(const torch::Tensor& w, bool transpose) {
+    // under hipGraph capture the repack ops must be RECORDED (the optimizer
+    // inside the graph mutates weights without a host-side version bump), so
+    // bypass the cache there
+    if (stream_capturing())
+        return transpose ? pack_fragment_major(w.t().contiguous())
+                         : pack_fragment_major(w);
+    auto& cache = transpose ? g_pack_cache_T : g_pack_cache;
+    const void* key = w.data_ptr();
+    auto it = cache.find(key);
+    if (it != cache.end() && it->second.version == g_weight_version &&
+        it->second.packed.numel() == w.numel())
+        return it->second.packed;
+    auto packed = transpose ? pack_fragment_major(w.t().contiguous())
+                            : pack_fragment_major(w);
+    cache[key] = {g_weight_version, packed};
+    return packed;
+}
+
 void check_dev_contig(const torch::Tensor& t, const char* name) {
     TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
     TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
@@ -152,7 +201,7 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
     TORCH_CHECK(h0.scalar_type() == dt && wh.scalar_type() == dt && c0.scalar_type() == dt,
                 "dtype mismatch");
 
-    auto wh_p = pack_fragment_major(wh);
+    auto wh_p = pack_cached(wh, /*transpose=*/false);
     auto h_seq = torch::empty({T, B, H}, gates_x.options());
     auto c_final = torch::empty({B, H}, gates_x.options());
     auto gates_act = torch::empty({stash ? T : 0, B, 4 * H}, gates_x.options());
@@ -211,8 +260,8 @@ lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
     const auto H = h00.size(1);
     TORCH_CHECK(gx1.size(1) == B && h01.size(1) == H, "pair shape mismatch");
     TORCH_CHECK(gx0.scalar_type() == torch::kBFloat16, "pair path is bf16-only");
-    auto wp0 = pack_fragment_major(wh0);
-    auto wp1 = pack_fragment_major(wh1);
+    auto wp0 = pack_cached(wh0, /*transpose=*/false);
+    auto wp1 = pack_cached(wh1, /*transpose=*/false);
     const int RT = (int)((B + 15) / 16);
     auto hx0 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
     auto hx1 = torch::zeros({2LL, RT, 16, H / 2}, gx0.options().dtype(torch::kInt64));
@@ -242,24 +291,27 @@ lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
 }
 
 // ---- K2: fused LSTM sequence backward -------------------------------------
+// Takes the UNTRANSPOSED weight wh [4H, H] (the parameter tensor itself) so
+// the transposed fragment-major pack can be cached across calls keyed on the
+// parameter's stable data_ptr (pack_cached).
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
 lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
              torch::Tensor gates_act, torch::Tensor c_seq, torch::Tensor c0,
-             torch::Tensor whT) {
+             torch::Tensor wh) {
     check_dev_contig(dh_seq, "dh_seq");
     check_dev_contig(dhT, "dhT");
     check_dev_contig(dcT, "dcT");
     check_dev_contig(gates_act, "gates_act");
     check_dev_contig(c_seq, "c_seq");
     check_dev_contig(c0, "c0");
-    check_dev_contig(whT, "whT");
+    check_dev_contig(wh, "wh");
     const auto T = dh_seq.size(0);
     const auto B = dh_seq.size(1);
     const auto H = dh_seq.size(2);
-    TORCH_CHECK(whT.size(0) == H && whT.size(1) == 4 * H, "whT must be [H, 4H]");
+    TORCH_CHECK(wh.size(0) == 4 * H && wh.size(1) == H, "wh must be [4H, H]");
     auto dt = dh_seq.scalar_type();
 
-    auto whT_p = pack_fragment_major(whT);
+    auto whT_p = pack_cached(wh, /*transpose=*/true);
     auto dgates = torch::empty({T, B, 4 * H}, dh_seq.options());
     auto dh0 = torch::empty({B, H}, dh_seq.options());
     auto dc0 = torch::empty({B, H}, dh_seq.options());
@@ -305,12 +357,12 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
            torch::Tensor, torch::Tensor, torch::Tensor>
 lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
                   torch::Tensor ga0, torch::Tensor cs0, torch::Tensor c00,
-                  torch::Tensor whT0,
+                  torch::Tensor wh0,
                   torch::Tensor dh1s, torch::Tensor dhT1, torch::Tensor dcT1,
                   torch::Tensor ga1, torch::Tensor cs1, torch::Tensor c01,
-                  torch::Tensor whT1) {
-    for (auto* t : {&dh0s, &dhT0, &dcT0, &ga0, &c00, &whT0,
-                    &dh1s, &dhT1, &dcT1, &ga1, &c01, &whT1})
+                  torch::Tensor wh1) {
+    for (auto* t : {&dh0s, &dhT0, &dcT0, &ga0, &c00, &wh0,
+                    &dh1s, &dhT1, &dcT1, &ga1, &c01, &wh1})
         check_dev_contig(*t, "bwd pair arg");
     const auto T0 = dh0s.size(0);
     const auto T1 = dh1s.size(0);
@@ -318,8 +370,9 @@ lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
     const auto H = dh0s.size(2);
     TORCH_CHECK(dh1s.size(1) == B && dh1s.size(2) == H, "bwd pair shape mismatch");
     TORCH_CHECK(dh0s.scalar_type() == torch::kBFloat16, "bwd pair is bf16-only");
-    auto wp0 = pack_fragment_major(whT0);
-    auto wp1 = pack_fragment_major(whT1);
+    TORCH_CHECK(wh0.size(0) == 4 * H && wh0.size(1) == H, "wh must be [4H, H]");
+    auto wp0 = pack_cached(wh0, /*transpose=*/true);
+    auto wp1 = pack_cached(wh1, /*transpose=*/true);
     const int RT = (int)((B + 15) / 16);
     auto opts = dh0s.options();
     auto dg0 = torch::empty({T0, B, 4 * H}, opts);
@@ -353,7 +406,8 @@ lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
 td_targets(torch::Tensor rewards, torch::Tensor dones, torch::Tensor q_boot,
            torch::Tensor q_online, double gamma, int64_t n_step, double eta,
-           bool rescale = false, double rescale_eps = 1e-3) {
+           bool rescale = false, double rescale_eps = 1e-3,
+           c10::optional<torch::Tensor> valid = c10::nullopt) {
     check_dev_contig(rewards, "rewards");
     check_dev_contig(dones, "dones");
     check_dev_contig(q_boot, "q_boot");
@@ -362,12 +416,19 @@ td_targets(torch::Tensor rewards, torch::Tensor dones, torch::Tensor q_boot,
     const auto T = q_online.size(1);
     const auto t_ext = rewards.size(1);
     TORCH_CHECK(rewards.scalar_type() == torch::kFloat32, "rewards must be fp32");
+    const int* valid_ptr = nullptr;
+    if (valid.has_value()) {
+        check_dev_contig(*valid, "valid");
+        TORCH_CHECK(valid->scalar_type() == torch::kInt32 && valid->size(0) == B,
+                    "valid must be int32 [B]");
+        valid_ptr = valid->data_ptr<int>();
+    }
     auto y = torch::empty({B, T}, q_online.options());
     auto td_abs = torch::empty({B, T}, q_online.options());
     auto prio = torch::empty({B}, q_online.options());
     launch_td_targets(rewards.data_ptr<float>(), dones.data_ptr<float>(),
                       q_boot.data_ptr<float>(), q_online.data_ptr<float>(),
-                      y.data_ptr<float>(), td_abs.data_ptr<float>(),
+                      valid_ptr, y.data_ptr<float>(), td_abs.data_ptr<float>(),
                       prio.data_ptr<float>(), (int)B, (int)T, (int)t_ext,
                       (int)n_step, (float)gamma, (float)eta,
                       rescale ? 1 : 0, (float)rescale_eps, cur_stream());
@@ -454,7 +515,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("td_targets", &td_targets, "fused n-step TD targets + priorities (K3)",
           py::arg("rewards"), py::arg("dones"), py::arg("q_boot"), py::arg("q_online"),
           py::arg("gamma"), py::arg("n_step"), py::arg("eta"),
-          py::arg("rescale") = false, py::arg("rescale_eps") = 1e-3);
+          py::arg("rescale") = false, py::arg("rescale_eps") = 1e-3,
+          py::arg("valid") = py::none());
     m.def("polyak_multi", &polyak_multi, "fused multi-tensor Polyak update (K4)");
     m.def("sumtree_sample", &sumtree_sample, "sum-tree stratified sample (K5)");
     m.def("sumtree_update", &sumtree_update, "sum-tree leaf update + propagate (K6)");
@@ -473,4 +535,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "two independent LSTM backwards in one launch (per-set T)");
     m.def("set_px_serialize", &set_px_serialize,
           "serialize column-split kernels device-wide (off by default)");
+    m.def("bump_weight_version", &bump_weight_version,
+          "invalidate the fragment-major weight-pack cache (call after every "
+          "optimizer/Polyak write)");
 }

@@ -11,11 +11,15 @@
 // delta vs online q; then wave-reduce max|d| and sum|d| -> priority.
 // Optional value-rescale: y = h(folded + g^n * mask * h_inv(q_boot)) applied
 // HERE so the priority/TD math matches the learner exactly.
+// valid (nullable, [B]): count of REAL (non-pad) training-window steps per
+// sequence — pad steps at the tail of flushed episodes contribute neither
+// td_abs nor priority mass (the learner masks them out of the loss too).
 extern "C" __global__ void __launch_bounds__(64)
 td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
                   const float* __restrict__ dones,    // [B, t_ext]
                   const float* __restrict__ q_boot,   // [B, T]
                   const float* __restrict__ q_online, // [B, T]
+                  const int* __restrict__ valid,      // [B] or nullptr
                   float* __restrict__ y_out,          // [B, T]
                   float* __restrict__ td_abs_out,     // [B, T]
                   float* __restrict__ prio_out,       // [B]
@@ -30,6 +34,11 @@ td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
     const float* qo_row = q_online + (long)b * T;
     float gamma_n = 1.0f;
     for (int k = 0; k < n_step; ++k) gamma_n *= gamma;
+    int vT = T;
+    if (valid) {
+        vT = valid[b];
+        vT = vT < 1 ? 1 : (vT > T ? T : vT);
+    }
 
     float max_d = 0.0f, sum_d = 0.0f;
     for (int t = lane; t < T; t += WAVE) {
@@ -53,7 +62,7 @@ td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
             float a = fabsf(y), s = y >= 0.f ? 1.f : -1.f;
             y = s * (sqrtf(a + 1.0f) - 1.0f) + rescale_eps * y;
         }
-        float d = fabsf(y - qo_row[t]);
+        float d = t < vT ? fabsf(y - qo_row[t]) : 0.0f;  // pad steps: no mass
         y_out[(long)b * T + t] = y;
         td_abs_out[(long)b * T + t] = d;
         max_d = fmaxf(max_d, d);
@@ -65,7 +74,7 @@ td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
         sum_d += __shfl_down(sum_d, off);
     }
     if (lane == 0)
-        prio_out[b] = eta * max_d + (1.0f - eta) * (sum_d / (float)T);
+        prio_out[b] = eta * max_d + (1.0f - eta) * (sum_d / (float)vT);
 }
 
 // --------------------------------------------------------------------- K4
@@ -109,11 +118,11 @@ __global__ void __launch_bounds__(256) polyak_multi_kernel(
 // ------------------------------------------------------------- launchers
 extern "C" void launch_td_targets(
     const float* rewards, const float* dones, const float* q_boot,
-    const float* q_online, float* y, float* td_abs, float* prio,
-    int B, int T, int t_ext, int n_step, float gamma, float eta,
+    const float* q_online, const int* valid, float* y, float* td_abs,
+    float* prio, int B, int T, int t_ext, int n_step, float gamma, float eta,
     int rescale, float rescale_eps, hipStream_t stream) {
     td_targets_kernel<<<dim3(B), dim3(WAVE), 0, stream>>>(
-        rewards, dones, q_boot, q_online, y, td_abs, prio,
+        rewards, dones, q_boot, q_online, valid, y, td_abs, prio,
         B, T, t_ext, n_step, gamma, eta, rescale, rescale_eps);
 }
 

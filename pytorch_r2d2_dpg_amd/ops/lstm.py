@@ -53,10 +53,9 @@ class _FusedLSTMSeq(torch.autograd.Function):
         H = h0.shape[1]
         ext = hip_ext()
         zero_hT = torch.zeros_like(h0)
-        whT = wh.t().contiguous()
         dgates, dh0, dc0 = ext.lstm_seq_bwd(
             dh_seq.contiguous(), zero_hT, dcT.contiguous(), gates_act, c_seq,
-            c0.contiguous(), whT,
+            c0.contiguous(), wh.contiguous(),
         )
         dg_flat = dgates.reshape(T * B, 4 * H)
         # library GEMMs for the dense gradients
@@ -146,10 +145,9 @@ class _FusedLSTMSeqPairHalf(torch.autograd.Function):
         H = h0.shape[1]
         ext = hip_ext()
         zero_hT = torch.zeros_like(h0)
-        whT = wh.t().contiguous()
         dgates, dh0, dc0 = ext.lstm_seq_bwd(
             dh_seq.contiguous(), zero_hT, dcT.contiguous(), gates_act, c_seq,
-            c0.contiguous(), whT,
+            c0.contiguous(), wh.contiguous(),
         )
         dg_flat = dgates.reshape(T * B, 4 * H)
         dx = (dg_flat @ wx).view(T, B, I)

@@ -34,6 +34,8 @@ def td_targets_and_priorities(
     eta: float,
     rescale: bool = False,        # R2D2 value rescaling h(.) (flag, default off)
     rescale_eps: float = 1e-3,
+    valid_n: torch.Tensor = None,  # [B] int32: real (non-pad) training steps;
+                                   # pad steps get td_abs=0 and no priority mass
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Returns (targets y [B,T], |delta| [B,T], priorities [B])."""
     T = q_online.shape[1]
@@ -42,6 +44,7 @@ def td_targets_and_priorities(
             rewards.contiguous(), dones.contiguous(), bootstrap_q.contiguous(),
             q_online.contiguous(), float(gamma), int(n_step), float(eta),
             rescale, rescale_eps,
+            valid_n.contiguous() if valid_n is not None else None,
         )
     qb = value_rescale_inv(bootstrap_q, rescale_eps) if rescale else bootstrap_q
     folded, mask = fold_nstep_ref(rewards, dones, gamma, n_step, T)
@@ -49,5 +52,13 @@ def td_targets_and_priorities(
     if rescale:
         y = value_rescale(y, rescale_eps)
     td_abs = (y - q_online).abs()
-    prio = sequence_priority_ref(td_abs, eta)
+    if valid_n is not None:
+        vn = valid_n.to(torch.long).clamp(1, T)
+        vmask = (torch.arange(T, device=td_abs.device).unsqueeze(0)
+                 < vn.unsqueeze(1)).to(td_abs.dtype)
+        td_abs = td_abs * vmask
+        prio = (eta * td_abs.max(dim=1).values
+                + (1.0 - eta) * td_abs.sum(dim=1) / vn.to(td_abs.dtype))
+    else:
+        prio = sequence_priority_ref(td_abs, eta)
     return y, td_abs, prio

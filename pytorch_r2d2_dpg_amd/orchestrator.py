@@ -28,15 +28,9 @@ from .replay.staging import StagingRing
 from .utils.metrics import MetricsLogger
 
 
-def _px_err() -> int:
-    try:
-        import torch as _t
-        if _t.cuda.is_available():
-            from .ops import hip_ext
-            return int(hip_ext().lstm_px_error())
-    except Exception:  # noqa: BLE001
-        pass
-    return 0
+# px spin-timeout surveillance lives in the LEARNER now (it polls the error
+# word every R2D2_PX_CHECK_EVERY steps and raises on corruption unless
+# R2D2_PX_ERR_WARN=1); the orchestrator just logs the accumulated counter.
 
 
 def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str] = None,
@@ -155,7 +149,7 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
                     eval_return=float(np.mean([r for _, r in pool.eval_returns[-10:]]))
                     if pool.eval_returns else 0.0,
                     actor_respawns=pool.respawn_count,
-                    px_err=_px_err(),
+                    px_err=learner.px_err_total,
                 )
                 t_last, trans_acc, env_steps_last = now, 0, env_steps
     finally:

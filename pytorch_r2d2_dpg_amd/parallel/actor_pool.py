@@ -195,7 +195,8 @@ class ActorPool:
                             ring.rew[lo:hi].numpy().copy(),
                             ring.done[lo:hi].numpy().copy(),
                             ring.state[lo:hi].numpy().copy(),
-                            ring.prio[lo:hi].numpy().copy()))
+                            ring.prio[lo:hi].numpy().copy(),
+                            ring.length[lo:hi].numpy().copy()))
                 self.total_sequences += hi - lo
             if ranges:
                 ring.commit()
@@ -205,10 +206,10 @@ class ActorPool:
         """Drain available sequences as objects (tests / small runs)."""
         out: List[Sequence] = []
         for batch in self.poll_arrays(max_batches):
-            obs, act, rew, done, state, prio = batch
+            obs, act, rew, done, state, prio, length = batch
             for i in range(obs.shape[0]):
                 out.append(Sequence(obs[i], act[i], rew[i], done[i], state[i],
-                                    float(prio[i]), obs.shape[1]))
+                                    float(prio[i]), int(length[i])))
         return out
 
     def ingest_rings(self, staging, max_per_ring: int = 0) -> int:

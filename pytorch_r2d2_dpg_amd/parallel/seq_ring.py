@@ -39,6 +39,7 @@ class SeqRing:
         self.done = shared((slots, L))
         self.state = shared((slots, 4, hidden_size))
         self.prio = shared((slots,))
+        self.length = shared((slots,), torch.int32)  # real (non-pad) steps
         self.wr = ctx.Value("l", 0)   # written-count (publish point)
         self.rd = ctx.Value("l", 0)   # consumed-count (backpressure)
         self.dropped = ctx.Value("l", 0)
@@ -61,6 +62,7 @@ class SeqRing:
         self.done[s].copy_(torch.from_numpy(seq.done))
         self.state[s].copy_(torch.from_numpy(seq.state))
         self.prio[s] = float(seq.priority)
+        self.length[s] = int(seq.length)
         with self.wr.get_lock():
             self.wr.value = wr + 1    # publish AFTER the slot is fully written
         return True

@@ -49,6 +49,9 @@ class SequenceReplay:
         self.reward = torch.zeros((capacity, L), dtype=torch.float32, device=self.device)
         self.done = torch.zeros((capacity, L), dtype=torch.float32, device=self.device)
         self.state = torch.zeros((capacity, 4, hidden_size), dtype=torch.float32, device=self.device)
+        self.length = torch.full((capacity,), stored_len, dtype=torch.int32,
+                                 device=self.device)
+        self.has_partial = False  # sticky: any padded sequence ingested
         self.tree = SumTree(capacity)
         self._next = 0
         self._size = 0
@@ -76,6 +79,10 @@ class SequenceReplay:
         self.reward[ti] = rew.to(self.device, non_blocking=True)
         self.done[ti] = done.to(self.device, non_blocking=True)
         self.state[ti] = state.to(self.device, non_blocking=True)
+        lengths = np.asarray([s.length for s in seqs], dtype=np.int32)
+        self.length[ti] = torch.from_numpy(lengths).to(self.device)
+        if (lengths < self.stored_len).any():
+            self.has_partial = True
         prios = np.asarray([max(s.priority, 1e-6) for s in seqs], dtype=np.float64)
         # publish-last: tree update makes the slots visible to sampling
         self.tree.update(idx, prios ** self.per_alpha)
@@ -100,6 +107,7 @@ class SequenceReplay:
             "done": self.done[ti],
             "state": self.state[ti],
             "weight": torch.from_numpy(w.astype(np.float32)).to(self.device),
+            "length": self.length[ti] if self.has_partial else None,
         }
 
     # ------------------------------------------------------- priority write

@@ -72,6 +72,12 @@ class GpuSequenceReplay:
         self.done = torch.zeros((capacity, L), dtype=torch.float32, device=self.device)
         self.state = torch.zeros((capacity, 4, hidden_size), dtype=store_dtype,
                                  device=self.device)
+        self.length = torch.full((capacity,), stored_len, dtype=torch.int32,
+                                 device=self.device)
+        # host-visible sticky flag: False until a padded (length < stored_len)
+        # sequence is ingested — lets sample() skip returning lengths and the
+        # learner skip the loss-mask entirely on full-length workloads (bench)
+        self.has_partial = False
         self.tree = torch.zeros(2 * self.cap2, dtype=torch.float32, device=self.device)
         self._next = 0
         self._size = 0
@@ -112,17 +118,27 @@ class GpuSequenceReplay:
         while int(pos[0]) < self.cap2:
             left = 2 * pos
             lmass = self.tree[left]
-            go_right = u >= lmass
+            # drift guard (mirrors sumtree.hip): fp error accumulating in
+            # "u -= lmass" can walk into a zero-mass right subtree spanning
+            # unwritten slots in [size, cap2) — fall back left there
+            go_right = (u >= lmass) & (self.tree[left + 1] > 0)
             u = torch.where(go_right, u - lmass, u)
             pos = left + go_right.long()
-        idx = pos - self.cap2
+        idx = (pos - self.cap2).clamp_(max=max(self._size - 1, 0))
         prob = self.tree[pos] / total
         return idx, prob
 
     # ------------------------------------------------------------------ add
     @torch.no_grad()
-    def add_batch(self, obs, action, reward, done, state, priority) -> None:
-        """Ingest a batch of sequences (torch tensors, any device)."""
+    def add_batch(self, obs, action, reward, done, state, priority,
+                  length=None, any_partial: bool = None) -> None:
+        """Ingest a batch of sequences (torch tensors, any device).
+
+        ``length`` [n] int = real (non-pad) steps per sequence (None = all
+        full). ``any_partial`` lets the caller pass a host-side precomputed
+        "any length < stored_len" so no device sync is needed; if omitted it
+        is derived from ``length`` (syncing when that tensor is on GPU).
+        """
         n = obs.shape[0]
         if n == 0:
             return
@@ -139,6 +155,17 @@ class GpuSequenceReplay:
         self.reward[didx] = reward.to(self.device, torch.float32, non_blocking=True)
         self.done[didx] = done.to(self.device, torch.float32, non_blocking=True)
         self.state[didx] = state.to(self.device, self.state.dtype, non_blocking=True)
+        if length is not None:
+            if not torch.is_tensor(length):
+                length = torch.as_tensor(np.asarray(length))
+            if any_partial is None:
+                src = length.cpu() if length.is_cuda else length
+                any_partial = bool((src < self.stored_len).any())
+            self.length[didx] = length.to(self.device, torch.int32, non_blocking=True)
+            if any_partial:
+                self.has_partial = True
+        else:
+            self.length[didx] = self.stored_len
         if not torch.is_tensor(priority):
             priority = torch.as_tensor(priority)
         p = priority.to(self.device, torch.float32).clamp_min(1e-6) ** self.per_alpha
@@ -171,6 +198,10 @@ class GpuSequenceReplay:
             "done": self.done.index_select(0, idx),
             "state": self.state.index_select(0, idx),
             "weight": w,
+            # only materialized once a padded sequence exists in the shard —
+            # the learner's masked-loss path keys off this being non-None
+            "length": (self.length.index_select(0, idx)
+                       if self.has_partial else None),
         }
 
     # ------------------------------------------------------- priority write

@@ -10,8 +10,15 @@ initial priority from the actor's own step-wise Q values:
     delta_t = fold_n(r)_t + gamma^n * mask_t * q_{t+n} - q_t
     p0      = eta * max|delta| + (1-eta) * mean|delta|
 
-At episode end the window is flushed (right-padded to fixed length with
-zero-reward terminal steps, mask excluded via done flags) and reset.
+At episode end the window is flushed (right-padded to fixed length by
+repeating the last observation with zero reward and done=0) and reset.
+Termination is carried solely by the REAL last step's done flag: after a
+terminal step the n-step fold's alive-mask is already 0, while after a
+time-limit truncation (done=0) the fold keeps bootstrapping through the
+repeated final observation — i.e. from the target net's value of the last
+real state — instead of wrongly treating truncation as termination. Padded
+steps are excluded from the critic loss and the priority reduction via
+``Sequence.length`` (carried through replay into the learner's mask).
 """
 
 from __future__ import annotations
@@ -93,7 +100,10 @@ class SequenceCutter:
         obs_arr = np.zeros((L,) + self.obs[0].shape, dtype=self.obs[0].dtype)
         act_arr = np.zeros((L, self.act[0].shape[0]), dtype=np.float32)
         rew_arr = np.zeros(L, dtype=np.float32)
-        done_arr = np.ones(L, dtype=np.float32)  # pad region counts as done
+        # pad region: done=0 (termination lives in the real flags; with the
+        # repeated-obs pad this bootstraps truncated episodes from the last
+        # real state's target value — see module docstring)
+        done_arr = np.zeros(L, dtype=np.float32)
         n = min(length, L)
         obs_arr[:n] = np.stack(self.obs[:n])
         act_arr[:n] = np.stack(self.act[:n])
@@ -106,7 +116,13 @@ class SequenceCutter:
                         self.states[0].copy(), float(prio), n)
 
     def _initial_priority(self, rew, done, q, n_real) -> float:
-        """n-step TD error against the actor's own step-wise Q estimates."""
+        """n-step TD error against the actor's own step-wise Q estimates.
+
+        Mirrors the learner's K3 semantics on the padded window: pad rewards
+        are zero, the alive-mask evolves from the REAL done flags only, and
+        the bootstrap beyond the real tail reads the last real Q (the
+        repeated-obs pad) — so truncated episodes bootstrap instead of being
+        treated as terminated."""
         t_lo = self.burn_in
         t_hi = max(t_lo + 1, min(self.seq_len, n_real))
         deltas = []
@@ -114,12 +130,11 @@ class SequenceCutter:
             acc, g, alive = 0.0, 1.0, 1.0
             for k in range(self.n_step):
                 if t + k >= n_real:
-                    alive = 0.0
-                    break
-                acc += g * rew[t + k]
+                    break  # pad rewards are zero, pad done is 0
+                acc += g * alive * rew[t + k]
                 alive *= 1.0 - done[t + k]
                 g *= self.gamma
-            boot = q[t + self.n_step] if (t + self.n_step) < n_real else 0.0
+            boot = q[t + self.n_step] if (t + self.n_step) < n_real else q[n_real - 1]
             y = acc + (self.gamma ** self.n_step) * alive * boot
             deltas.append(abs(y - q[t]) if t < n_real else 0.0)
         d = np.asarray(deltas, dtype=np.float64)

@@ -44,6 +44,7 @@ class StagingRing:
         self._done = mk((L,))
         self._state = mk((4, H))
         self._prio = mk(())
+        self._len = mk((), torch.int32)  # real (non-pad) steps per sequence
         self._events = [torch.cuda.Event() if pin else None for _ in range(slots)]
         self._busy = [False] * slots
         self._cursor = 0
@@ -87,6 +88,7 @@ class StagingRing:
                     self._done[s][fill:fill + n].copy_(ring.done[off:off + n])
                     self._state[s][fill:fill + n].copy_(ring.state[off:off + n])
                     self._prio[s][fill:fill + n].copy_(ring.prio[off:off + n])
+                    self._len[s][fill:fill + n].copy_(ring.length[off:off + n])
                     fill += n
                     off += n
                     if fill == self.slot_seqs:
@@ -99,10 +101,11 @@ class StagingRing:
         return total
 
     @torch.no_grad()
-    def ingest_arrays(self, obs, act, rew, done, state, prio) -> int:
+    def ingest_arrays(self, obs, act, rew, done, state, prio, length=None) -> int:
         """Stage stacked numpy arrays (the fast path from poll_arrays)."""
         total = 0
         n_all = obs.shape[0]
+        L = self.replay.stored_len
         for off in range(0, n_all, self.slot_seqs):
             n = min(self.slot_seqs, n_all - off)
             s = self._next_slot()
@@ -112,6 +115,10 @@ class StagingRing:
             self._done[s][:n].copy_(torch.from_numpy(done[off:off + n]))
             self._state[s][:n].copy_(torch.from_numpy(state[off:off + n]))
             self._prio[s][:n].copy_(torch.from_numpy(prio[off:off + n]))
+            if length is None:
+                self._len[s][:n].fill_(L)
+            else:
+                self._len[s][:n].copy_(torch.from_numpy(length[off:off + n]).to(torch.int32))
             total += self._dispatch(s, n)
         return total
 
@@ -131,17 +138,23 @@ class StagingRing:
             self._done[s][:n].copy_(torch.from_numpy(np.stack([q.done for q in chunk])))
             self._state[s][:n].copy_(torch.from_numpy(np.stack([q.state for q in chunk])))
             self._prio[s][:n].copy_(torch.tensor([q.priority for q in chunk]))
+            self._len[s][:n].copy_(torch.tensor([q.length for q in chunk],
+                                                dtype=torch.int32))
             total += self._dispatch(s, n)
         return total
 
     @torch.no_grad()
     def _dispatch(self, s: int, n: int) -> int:
+        # host-side partial check (pinned buffer, pre-copy): lets the replay
+        # skip the loss-mask path entirely until a padded sequence exists
+        any_partial = bool((self._len[s][:n] < self.replay.stored_len).any())
         if self.is_cuda:
             # bulk H2D on the side stream, scatter on the learner stream
             with torch.cuda.stream(self.stream):
                 dev = [t[:n].to(self.device, non_blocking=True)
                        for t in (self._obs[s], self._act[s], self._rew[s],
-                                 self._done[s], self._state[s], self._prio[s])]
+                                 self._done[s], self._state[s], self._prio[s],
+                                 self._len[s])]
                 self._events[s].record(self.stream)
             self._busy[s] = True
             cur = torch.cuda.current_stream()
@@ -153,9 +166,10 @@ class StagingRing:
             # ingest — silent replay corruption)
             for t in dev:
                 t.record_stream(cur)
-            self.replay.add_batch(*dev)
+            self.replay.add_batch(*dev[:6], length=dev[6], any_partial=any_partial)
         else:
             self.replay.add_batch(self._obs[s][:n], self._act[s][:n],
                                   self._rew[s][:n], self._done[s][:n],
-                                  self._state[s][:n], self._prio[s][:n])
+                                  self._state[s][:n], self._prio[s][:n],
+                                  length=self._len[s][:n], any_partial=any_partial)
         return n

@@ -35,10 +35,12 @@ def test_cutter_invariants_any_episode_structure(dones):
         # non-pad region is consecutive env steps
         for k in range(s.length):
             assert s.obs[k, 0] == t0 + k
-        # pad region repeats the last real obs and is marked done
+        # pad region repeats the last real obs with done=0 (termination is
+        # carried by the REAL last step's flag; truncation bootstraps
+        # through the repeated obs)
         for k in range(s.length, STORED):
             assert s.obs[k, 0] == s.obs[s.length - 1, 0]
-            assert s.done[k] == 1.0
+            assert s.done[k] == 0.0
         assert np.isfinite(s.priority) and s.priority >= 0.0
     # every emitted window lies inside one episode (no done strictly inside
     # the real region except at its last step)

@@ -121,7 +121,8 @@ def test_cutter_flush_on_done_pads():
     s = emitted[0]
     assert s.length == 7
     assert s.obs.shape[0] == 10
-    assert s.done[6] == 1.0 and s.done[7] == 1.0  # pad marked done
+    assert s.done[6] == 1.0  # the REAL terminal flag
+    assert s.done[7] == 0.0  # pad done=0 (masked out via s.length instead)
     assert s.obs[8, 0] == 6.0  # pad repeats last obs
     # window reset after done
     assert len(c.obs) == 0

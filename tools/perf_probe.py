@@ -61,7 +61,7 @@ def main():
 
     h_seq, c_final, gates_act, c_seq = ext.lstm_seq_fwd(gates_x, h0, c0, wh)
     dh = torch.randn_like(h_seq)
-    whT = wh.t().contiguous()
+    whT = wh.contiguous()  # lstm_seq_bwd takes the untransposed weight now
     zero = torch.zeros_like(h0)
     t_bwd_k = timeit(lambda: ext.lstm_seq_bwd(dh, zero, zero, gates_act, c_seq, c0, whT))
     print(f"K2 lstm bwd kernel only      : {t_bwd_k:8.3f} ms")

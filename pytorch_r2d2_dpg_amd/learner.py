@@ -125,15 +125,14 @@ class Learner:
         # chains, +20% measured); at larger batches two concurrent
         # column-split grids contend for CU residency and crawl, so run the
         # chains sequentially there (the px kernels fill the chip anyway).
-        # Also gated on train_len <= 48: at B=256 the two concurrent bwd-px
-        # grids are an exact 256-CU fit, and with the longer kernels of
-        # seq-120 configs the collision window with stray kernels (ingest
-        # scatter, GEMMs) is wide enough that displaced blocks spin out
-        # (observed as multi-second steps); sequential chains are safe and
-        # near-parity there. R2D2_NO_OVERLAP=1 forces off.
+        # The r1 train_len <= 48 gate (seq-120 spin-outs when two bwd-px
+        # grids were an exact 256-CU fit) is GONE: the px launchers now do
+        # host-side CU-budget accounting (bindings.cpp px_budget_*) — a
+        # launch that would over-subscribe the chip picks a smaller CG or
+        # serializes behind the in-flight grid, so concurrent px can no
+        # longer livelock. R2D2_NO_OVERLAP=1 forces off.
         overlap_ok = (self.device.type == "cuda"
                       and cfg.learner.batch_size <= 256
-                      and self.train_len <= 48
                       and not os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()
@@ -560,24 +559,31 @@ class Learner:
         assert self.cfg.learner.target_update_every == 1
         # the paired-backward path uses record_stream (allocator hints),
         # which is not legal inside graph capture — capture the standard
-        # autograd path instead (identical math)
+        # autograd path instead (identical math). Restored on FAILURE so an
+        # aborted capture doesn't silently change the eager path selection.
+        pair_saved = self._pair_bwd
         self._pair_bwd = False
-        keys = ("obs", "action", "reward", "done", "state", "weight", "idx")
-        self._static = {k: example_batch[k].clone() for k in keys}
-        # capturable Adam: state must live on device, step as device tensor
-        for opt in (self.opt_actor, self.opt_critic):
-            for gparam in opt.param_groups:
-                gparam["capturable"] = True
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for _ in range(3):  # warmup allocations + optimizer state
-                self._step_core(self._static, zero_to_none=False)
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
-            self._graph_stats = self._step_core(self._static, zero_to_none=False)
+        try:
+            keys = ("obs", "action", "reward", "done", "state", "weight", "idx")
+            self._static = {k: example_batch[k].clone() for k in keys}
+            # capturable Adam: state must live on device, step as device tensor
+            for opt in (self.opt_actor, self.opt_critic):
+                for gparam in opt.param_groups:
+                    gparam["capturable"] = True
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):  # warmup allocations + optimizer state
+                    self._step_core(self._static, zero_to_none=False)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._graph_stats = self._step_core(self._static, zero_to_none=False)
+        except Exception:
+            self._pair_bwd = pair_saved
+            self._graph = None
+            raise
 
     # ------------------------------------------------------------ train step
     def train_step(self, batch: Dict[str, torch.Tensor]) -> StepStats:

@@ -38,24 +38,24 @@ extern "C" void launch_fp8_gather_decode(const unsigned char*, const long*, void
                                          long, int, float, hipStream_t);
 extern "C" int launch_lstm_fwd_px_bf16(const void*, const void*, const void*, const void*,
                                        void*, void*, void*, float*, unsigned long long*,
-                                       unsigned int*, int, int, int, int, hipStream_t);
+                                       unsigned int*, int, int, int, int, int, hipStream_t);
 extern "C" int launch_lstm_fwd_px2_bf16(const void*, const void*, const void*, const void*,
                                         void*, void*, void*, float*, int, int,
                                         unsigned long long*,
                                         const void*, const void*, const void*, const void*,
                                         void*, void*, void*, float*, int, int,
                                         unsigned long long*,
-                                        unsigned int*, int, int, hipStream_t);
+                                        unsigned int*, int, int, int, hipStream_t);
 extern "C" int launch_lstm_bwd_px_bf16(const void*, const void*, const void*, const void*,
                                        const float*, const void*, const void*, void*, void*,
                                        void*, unsigned char*, unsigned long long*,
-                                       unsigned int*, int, int, int, hipStream_t);
+                                       unsigned int*, int, int, int, int, hipStream_t);
 extern "C" int launch_lstm_bwd_px2_bf16(
     const void*, const void*, const void*, const void*, const float*, const void*,
     const void*, void*, void*, void*, unsigned char*, unsigned long long*, int,
     const void*, const void*, const void*, const void*, const float*, const void*,
     const void*, void*, void*, void*, unsigned char*, unsigned long long*, int,
-    unsigned int*, int, int, hipStream_t);
+    unsigned int*, int, int, int, hipStream_t);
 
 namespace {
 
@@ -92,6 +92,59 @@ void px_chain_record(hipStream_t st) {
 }
 
 void set_px_serialize(bool on) { g_px_serialize = on; }
+
+// ---- CU-budget accounting for column-split grids ---------------------------
+// Every px kernel requires ALL its blocks co-resident (cross-block spins), so
+// the blocks of all CONCURRENTLY-RUNNING px grids must sum to <= 224 of the
+// 256 CUs. The host tracks enqueued-and-possibly-running px launches via
+// events; each new launch gets the remaining budget and the launcher picks a
+// smaller CG (fewer blocks) under contention. If nothing fits, the caller's
+// stream first waits on every in-flight px event (back-to-back px is fast;
+// CONCURRENT over-subscribed px livelocks until the spin bound). GIL
+// serializes all access. R2D2_PX_NO_BUDGET=1 restores the fixed 224 limit.
+struct PxInflight { hipEvent_t ev; int blocks; };
+std::vector<PxInflight> g_px_inflight;       // current device only (per-proc 1 GPU)
+std::vector<hipEvent_t> g_px_event_pool;
+bool g_px_budget_enabled = []() {
+    const char* e = getenv("R2D2_PX_NO_BUDGET");
+    return !(e && e[0] && e[0] != '0');
+}();
+
+int px_budget_reserve() {
+    if (!g_px_budget_enabled) return 224;
+    int used = 0;
+    for (size_t i = 0; i < g_px_inflight.size();) {
+        if (hipEventQuery(g_px_inflight[i].ev) == hipSuccess) {
+            g_px_event_pool.push_back(g_px_inflight[i].ev);
+            g_px_inflight[i] = g_px_inflight.back();
+            g_px_inflight.pop_back();
+        } else {
+            used += g_px_inflight[i].blocks;
+            ++i;
+        }
+    }
+    return 224 - used;
+}
+
+// make `st` wait for every in-flight px grid, then return the full budget
+int px_budget_drain(hipStream_t st) {
+    for (auto& f : g_px_inflight)
+        (void)hipStreamWaitEvent(st, f.ev, 0);
+    return 224;
+}
+
+void px_budget_commit(int blocks, hipStream_t st) {
+    if (!g_px_budget_enabled) return;
+    hipEvent_t ev;
+    if (!g_px_event_pool.empty()) {
+        ev = g_px_event_pool.back();
+        g_px_event_pool.pop_back();
+    } else {
+        (void)hipEventCreateWithFlags(&ev, hipEventDisableTiming);
+    }
+    (void)hipEventRecord(ev, st);
+    g_px_inflight.push_back({ev, blocks});
+}
 
 // Per-device error word for the column-split kernel's bounded spins.
 torch::Tensor& px_err_word(const torch::Device& dev) {
@@ -217,14 +270,24 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
                                gates_x.options().dtype(torch::kInt64));
         auto& err = px_err_word(gates_x.device());
         px_chain_wait(cur_stream());
-        rc = launch_lstm_fwd_px_bf16(
-            gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(), wh_p.data_ptr(),
-            h_seq.data_ptr(), c_final.data_ptr(), gates_act.data_ptr(),
-            stash ? c_seq.data_ptr<float>() : nullptr,
-            (unsigned long long*)hx.data_ptr<int64_t>(),
-            (unsigned int*)err.data_ptr<int32_t>(),
-            (int)T, (int)B, (int)H, stash ? 1 : 0, cur_stream());
-        if (rc == 0) px_chain_record(cur_stream());
+        int budget = px_budget_reserve();
+        auto try_px = [&](int bud) {
+            return launch_lstm_fwd_px_bf16(
+                gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(), wh_p.data_ptr(),
+                h_seq.data_ptr(), c_final.data_ptr(), gates_act.data_ptr(),
+                stash ? c_seq.data_ptr<float>() : nullptr,
+                (unsigned long long*)hx.data_ptr<int64_t>(),
+                (unsigned int*)err.data_ptr<int32_t>(),
+                (int)T, (int)B, (int)H, stash ? 1 : 0, bud, cur_stream());
+        };
+        rc = try_px(budget);
+        if (rc < 0 && budget < 224)  // contended: serialize behind in-flight px
+            rc = try_px(px_budget_drain(cur_stream()));
+        if (rc > 0) {
+            px_budget_commit(rc, cur_stream());
+            px_chain_record(cur_stream());
+            rc = 0;
+        }
         if (rc != 0)
             rc = launch_lstm_fwd_bf16(gates_x.data_ptr(), h0.data_ptr(), c0.data_ptr(),
                                       wh_p.data_ptr(), h_seq.data_ptr(), c_final.data_ptr(),
@@ -276,17 +339,25 @@ lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
     auto cs1 = torch::empty({stash1 ? T1 : 0, B, H},
                             gx0.options().dtype(torch::kFloat32));
     auto& err = px_err_word(gx0.device());
-    int rc = launch_lstm_fwd_px2_bf16(
-        gx0.data_ptr(), h00.data_ptr(), c00.data_ptr(), wp0.data_ptr(),
-        hs0.data_ptr(), cf0.data_ptr(), ga0.data_ptr(),
-        stash0 ? cs0.data_ptr<float>() : nullptr, stash0 ? 1 : 0, (int)T0,
-        (unsigned long long*)hx0.data_ptr<int64_t>(),
-        gx1.data_ptr(), h01.data_ptr(), c01.data_ptr(), wp1.data_ptr(),
-        hs1.data_ptr(), cf1.data_ptr(), ga1.data_ptr(),
-        stash1 ? cs1.data_ptr<float>() : nullptr, stash1 ? 1 : 0, (int)T1,
-        (unsigned long long*)hx1.data_ptr<int64_t>(),
-        (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, cur_stream());
-    TORCH_CHECK(rc == 0, "pair geometry unsupported (fall back to two calls)");
+    int budget = px_budget_reserve();
+    auto try_px = [&](int bud) {
+        return launch_lstm_fwd_px2_bf16(
+            gx0.data_ptr(), h00.data_ptr(), c00.data_ptr(), wp0.data_ptr(),
+            hs0.data_ptr(), cf0.data_ptr(), ga0.data_ptr(),
+            stash0 ? cs0.data_ptr<float>() : nullptr, stash0 ? 1 : 0, (int)T0,
+            (unsigned long long*)hx0.data_ptr<int64_t>(),
+            gx1.data_ptr(), h01.data_ptr(), c01.data_ptr(), wp1.data_ptr(),
+            hs1.data_ptr(), cf1.data_ptr(), ga1.data_ptr(),
+            stash1 ? cs1.data_ptr<float>() : nullptr, stash1 ? 1 : 0, (int)T1,
+            (unsigned long long*)hx1.data_ptr<int64_t>(),
+            (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, bud,
+            cur_stream());
+    };
+    int rc = try_px(budget);
+    if (rc < 0 && budget < 224)
+        rc = try_px(px_budget_drain(cur_stream()));
+    TORCH_CHECK(rc > 0, "pair geometry unsupported (fall back to two calls)");
+    px_budget_commit(rc, cur_stream());
     return {hs0, cf0, ga0, cs0, hs1, cf1, ga1, cs1};
 }
 
@@ -324,14 +395,24 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
         auto flags = torch::zeros({2LL, RT, 8}, dh_seq.options().dtype(torch::kInt64));
         auto& err = px_err_word(dh_seq.device());
         px_chain_wait(cur_stream());
-        rc = launch_lstm_bwd_px_bf16(
-            dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(), gates_act.data_ptr(),
-            c_seq.data_ptr<float>(), c0.data_ptr(), whT_p.data_ptr(),
-            dgates.data_ptr(), dh0.data_ptr(), dc0.data_ptr(),
-            slab.data_ptr<uint8_t>(), (unsigned long long*)flags.data_ptr<int64_t>(),
-            (unsigned int*)err.data_ptr<int32_t>(),
-            (int)T, (int)B, (int)H, cur_stream());
-        if (rc == 0) px_chain_record(cur_stream());
+        int budget = px_budget_reserve();
+        auto try_px = [&](int bud) {
+            return launch_lstm_bwd_px_bf16(
+                dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(), gates_act.data_ptr(),
+                c_seq.data_ptr<float>(), c0.data_ptr(), whT_p.data_ptr(),
+                dgates.data_ptr(), dh0.data_ptr(), dc0.data_ptr(),
+                slab.data_ptr<uint8_t>(), (unsigned long long*)flags.data_ptr<int64_t>(),
+                (unsigned int*)err.data_ptr<int32_t>(),
+                (int)T, (int)B, (int)H, bud, cur_stream());
+        };
+        rc = try_px(budget);
+        if (rc < 0 && budget < 224)
+            rc = try_px(px_budget_drain(cur_stream()));
+        if (rc > 0) {
+            px_budget_commit(rc, cur_stream());
+            px_chain_record(cur_stream());
+            rc = 0;
+        }
         if (rc != 0)
             rc = launch_lstm_bwd_bf16(dh_seq.data_ptr(), dhT.data_ptr(), dcT.data_ptr(),
                                       gates_act.data_ptr(), c_seq.data_ptr<float>(),
@@ -386,19 +467,27 @@ lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
     auto flags0 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
     auto flags1 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
     auto& err = px_err_word(dh0s.device());
-    int rc = launch_lstm_bwd_px2_bf16(
-        dh0s.data_ptr(), dhT0.data_ptr(), dcT0.data_ptr(), ga0.data_ptr(),
-        cs0.data_ptr<float>(), c00.data_ptr(), wp0.data_ptr(),
-        dg0.data_ptr(), dh00.data_ptr(), dc00.data_ptr(),
-        slab0.data_ptr<uint8_t>(), (unsigned long long*)flags0.data_ptr<int64_t>(),
-        (int)T0,
-        dh1s.data_ptr(), dhT1.data_ptr(), dcT1.data_ptr(), ga1.data_ptr(),
-        cs1.data_ptr<float>(), c01.data_ptr(), wp1.data_ptr(),
-        dg1.data_ptr(), dh01.data_ptr(), dc01.data_ptr(),
-        slab1.data_ptr<uint8_t>(), (unsigned long long*)flags1.data_ptr<int64_t>(),
-        (int)T1,
-        (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, cur_stream());
-    TORCH_CHECK(rc == 0, "bwd pair geometry unsupported (fall back to two calls)");
+    int budget = px_budget_reserve();
+    auto try_px = [&](int bud) {
+        return launch_lstm_bwd_px2_bf16(
+            dh0s.data_ptr(), dhT0.data_ptr(), dcT0.data_ptr(), ga0.data_ptr(),
+            cs0.data_ptr<float>(), c00.data_ptr(), wp0.data_ptr(),
+            dg0.data_ptr(), dh00.data_ptr(), dc00.data_ptr(),
+            slab0.data_ptr<uint8_t>(), (unsigned long long*)flags0.data_ptr<int64_t>(),
+            (int)T0,
+            dh1s.data_ptr(), dhT1.data_ptr(), dcT1.data_ptr(), ga1.data_ptr(),
+            cs1.data_ptr<float>(), c01.data_ptr(), wp1.data_ptr(),
+            dg1.data_ptr(), dh01.data_ptr(), dc01.data_ptr(),
+            slab1.data_ptr<uint8_t>(), (unsigned long long*)flags1.data_ptr<int64_t>(),
+            (int)T1,
+            (unsigned int*)err.data_ptr<int32_t>(), (int)B, (int)H, bud,
+            cur_stream());
+    };
+    int rc = try_px(budget);
+    if (rc < 0 && budget < 224)
+        rc = try_px(px_budget_drain(cur_stream()));
+    TORCH_CHECK(rc > 0, "bwd pair geometry unsupported (fall back to two calls)");
+    px_budget_commit(rc, cur_stream());
     return {dg0, dh00, dc00, dg1, dh01, dc01};
 }
 

@@ -221,20 +221,26 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     }
 }
 
+// All px launchers take a CU budget (<= 224) from the caller's in-flight
+// accounting (bindings.cpp): every block of a px grid must be co-resident
+// with all its peers, so the sum of concurrently-running px blocks must
+// never exceed the chip. Success returns the BLOCK COUNT (positive) so the
+// caller can commit it against the budget; failure returns -1.
 extern "C" int launch_lstm_fwd_px_bf16(
     const void* gates_x, const void* h0, const void* c0, const void* wh_packed,
     void* h_seq, void* c_final, void* gates_act, float* c_seq,
     unsigned long long* hx, unsigned int* err,
-    int T, int B, int H, int stash, hipStream_t stream) {
+    int T, int B, int H, int stash, int budget, hipStream_t stream) {
     int RT = (B + 15) / 16;
     // largest CG with co-residency margin and >= 2 fragments per gate
     if (H & (H - 1)) return -1;              // sweep indexing needs pow2 H
+    if (budget > 224) budget = 224;
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;          // wave owns 16 cols
         int w = (H / c) / 16;                // waves per block
         if (w < 2 || w > 16) continue;
-        if (RT * c > 224) continue;          // all blocks must be co-resident
+        if (RT * c > budget) continue;       // all blocks must be co-resident
         CG = c;
         break;
     }
@@ -268,7 +274,7 @@ extern "C" int launch_lstm_fwd_px_bf16(
             return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return 0;
+    return RT * CG;
 }
 
 // ---------------------------------------------------------------- backward
@@ -481,19 +487,29 @@ DEV_INLINE void lstm_bwd_px_body(const BwdPxArgs& a, unsigned int* err,
         }
         __syncthreads();
 
-        // ---- dh recurrence: dh_prev(own cols) = dgates(all 4H) @ Wh
-        f32x4 acc = {0, 0, 0, 0};
-        for (int kk = 0; kk < H4; kk += 32) {
-            int k = kk + a_kgrp;
-            short8 a_frag = *reinterpret_cast<const short8*>(
-                &buf[px_swz(a_row * rowbytes4 + k * 2, a_row)]);
-            int nb = (col_base >> 4);
-            long off = (((long)nb * (H4 >> 5) + (kk >> 5)) * 64 + lane) * 8;
-            short8 b_frag = *reinterpret_cast<const short8*>(&whT[off]);
-            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
+        // ---- dh recurrence: dh_prev(own cols) = dgates(all 4H) @ Wh.
+        // K = 4H with ONE output fragment: a single accumulator would chain
+        // 4H/32 serially-dependent MFMAs (latency-bound at the 2 waves/SIMD
+        // this kernel runs at) — split K over 4 accumulators for ILP, sum at
+        // the end (unlike the fwd, whose 4 gate fragments give ILP for free)
+        f32x4 acc4[4] = {};
+        const int nb = (col_base >> 4);
+        for (int kk = 0; kk < H4; kk += 128) {
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                int kx = kk + u * 32;
+                int k = kx + a_kgrp;
+                short8 a_frag = *reinterpret_cast<const short8*>(
+                    &buf[px_swz(a_row * rowbytes4 + k * 2, a_row)]);
+                long off = (((long)nb * (H4 >> 5) + (kx >> 5)) * 64 + lane) * 8;
+                short8 b_frag = *reinterpret_cast<const short8*>(&whT[off]);
+                acc4[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, b_frag, acc4[u], 0, 0, 0);
+            }
         }
 #pragma unroll
-        for (int r = 0; r < 4; ++r) dh_reg[r] = acc[r];
+        for (int r = 0; r < 4; ++r)
+            dh_reg[r] = acc4[0][r] + acc4[1][r] + acc4[2][r] + acc4[3][r];
         __syncthreads();  // MFMA reads done before next step's LDS writes
     }
 #pragma unroll
@@ -532,10 +548,11 @@ extern "C" int launch_lstm_bwd_px_bf16(
     const float* c_seq, const void* c0, const void* whT_packed,
     void* dgates, void* dh0, void* dc0,
     unsigned char* slab, unsigned long long* flags, unsigned int* err,
-    int T, int B, int H, hipStream_t stream) {
+    int T, int B, int H, int budget, hipStream_t stream) {
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
     if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;  // keep 2 blocks/CU
+    if (budget > 224) budget = 224;
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
@@ -545,7 +562,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
                                    // swizzle permutes 16B blocks within a
                                    // 128B window, so raw byte-range copies
                                    // are closed only for >=128B regions
-        if (RT * c > 224) continue;
+        if (RT * c > budget) continue;
         CG = c;
         break;
     }
@@ -565,7 +582,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return 0;
+    return RT * CG;
 }
 
 extern "C" int launch_lstm_bwd_px2_bf16(
@@ -575,17 +592,18 @@ extern "C" int launch_lstm_bwd_px2_bf16(
     const void* dh_seq1, const void* dhT1, const void* dcT1, const void* ga1,
     const float* cs1, const void* c01, const void* whT1, void* dg1, void* dh01,
     void* dc01, unsigned char* slab1, unsigned long long* flags1, int T1,
-    unsigned int* err, int B, int H, hipStream_t stream) {
+    unsigned int* err, int B, int H, int budget, hipStream_t stream) {
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
     if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;
+    if (budget > 224) budget = 224;
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
         if (H / c < 64) continue;        // slab swizzle closure (see solo)
-        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
+        if (RT * c * 2 > budget) continue;  // BOTH halves must co-reside
         CG = c;
         break;
     }
@@ -610,7 +628,7 @@ extern "C" int launch_lstm_bwd_px2_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return 0;
+    return RT * CG * 2;
 }
 
 // ------------------------------------------------- paired (z=2) forward
@@ -804,15 +822,16 @@ extern "C" int launch_lstm_fwd_px2_bf16(
     const void* gx1, const void* h01, const void* c01, const void* wh1,
     void* hs1, void* cf1, void* ga1, float* cs1, int stash1, int T1,
     unsigned long long* hx1,
-    unsigned int* err, int B, int H, hipStream_t stream) {
+    unsigned int* err, int B, int H, int budget, hipStream_t stream) {
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
+    if (budget > 224) budget = 224;
     int CG = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
-        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
+        if (RT * c * 2 > budget) continue;  // BOTH halves must co-reside
         CG = c;
         break;
     }
@@ -835,5 +854,5 @@ extern "C" int launch_lstm_fwd_px2_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return 0;
+    return RT * CG * 2;
 }

@@ -77,10 +77,44 @@ extern "C" void launch_sumtree_sample(const float* tree, int cap, int B,
         tree, cap, B, seed, step, idx_out, prob_out);
 }
 
+// Multi-block variant for large update batches: one kernel launch per level
+// (the launch boundary IS the level barrier). Within a level, duplicate
+// parents get identical recomputed sums and children were finalized by the
+// previous launch — benign, deterministic, no atomics.
+extern "C" __global__ void __launch_bounds__(256)
+sumtree_update_leaves_kernel(float* __restrict__ tree, int cap,
+                             const long* __restrict__ idx,
+                             const float* __restrict__ prio, int M) {
+    int m = blockIdx.x * (int)blockDim.x + threadIdx.x;
+    if (m < M) tree[cap + idx[m]] = prio[m];
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+sumtree_update_level_kernel(float* __restrict__ tree, int cap,
+                            const long* __restrict__ idx, int M, int level) {
+    int m = blockIdx.x * (int)blockDim.x + threadIdx.x;
+    if (m < M) {
+        int pos = (int)((cap + idx[m]) >> level);
+        tree[pos] = tree[2 * pos] + tree[2 * pos + 1];
+    }
+}
+
 extern "C" void launch_sumtree_update(float* tree, int cap, const long* idx,
                                       const float* prio, int M, hipStream_t stream) {
     int depth = 0;
     while ((1 << depth) < cap) ++depth;
-    // leaves sit at depth `depth`; root is reached after `depth` halvings
-    sumtree_update_kernel<<<dim3(1), dim3(1024), 0, stream>>>(tree, cap, idx, prio, M, depth);
+    if (M <= 4096) {
+        // leaves sit at depth `depth`; root is reached after `depth` halvings
+        sumtree_update_kernel<<<dim3(1), dim3(1024), 0, stream>>>(
+            tree, cap, idx, prio, M, depth);
+        return;
+    }
+    // large batch (bulk ingest, 8-rank rates): one grid-stride launch per
+    // level so all CUs participate; the stream orders the levels
+    int blocks = (M + 255) / 256;
+    sumtree_update_leaves_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+        tree, cap, idx, prio, M);
+    for (int level = 1; level <= depth; ++level)
+        sumtree_update_level_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+            tree, cap, idx, M, level);
 }

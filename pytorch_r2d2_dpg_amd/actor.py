@@ -91,13 +91,16 @@ class ActorWorker:
                 r += rk
                 if done or trunc:
                     break
+            # timeout_terminal: treat the time limit as termination (config
+            # choice — see ReplayConfig); else PEB (bootstrap through pad)
+            d_eff = done or (trunc and self.cfg.replay.timeout_terminal)
             self.episode_return += r
             self.env_steps += 1
             out.extend(self.cutter.push(self._obs, a, r * self.cfg.replay.reward_scale,
-                                        done, float(q[0]), snapshot))
+                                        d_eff, float(q[0]), snapshot))
             self._obs = next_obs
             if done or trunc:
-                if trunc and not done:
+                if trunc and not d_eff:
                     out.extend(self.cutter.flush())
                 self.episode_returns.append(self.episode_return)
                 self.episode_return = 0.0
@@ -183,6 +186,7 @@ class VectorActorWorker:
             q, (self._h_c, self._c_c) = self.critic_net.step(
                 feat, torch.from_numpy(a), (self._h_c, self._c_c))
             q_np = q.numpy()
+            tt = self.cfg.replay.timeout_terminal
             for e, env in enumerate(self.envs):
                 r_acc, done, trunc = 0.0, False, False
                 for _ in range(rep):
@@ -190,12 +194,13 @@ class VectorActorWorker:
                     r_acc += r
                     if done or trunc:
                         break
+                d_eff = done or (trunc and tt)
                 self.episode_return[e] += r_acc
                 out.extend(self.cutters[e].push(self._obs[e], a[e], r_acc * rs,
-                                                done, float(q_np[e]), snap[e]))
+                                                d_eff, float(q_np[e]), snap[e]))
                 self._obs[e] = next_obs
                 if done or trunc:
-                    if trunc and not done:
+                    if trunc and not d_eff:
                         out.extend(self.cutters[e].flush())
                     self.episode_returns.append(float(self.episode_return[e]))
                     self.episode_return[e] = 0.0

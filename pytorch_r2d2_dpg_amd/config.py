@@ -55,6 +55,16 @@ class ReplayConfig:
     min_fill: int = 256             # sequences before learning starts
     reward_scale: float = 1.0       # rewards are scaled at actor-push time
     fp8_obs: bool = False           # store observations as fp8 e4m3 in the GPU shard
+    # Time-limit handling. False (default) = partial-episode bootstrapping:
+    # truncation keeps done=0 and the learner bootstraps from the target
+    # net's value of the last real state (via the repeated-obs pad) — the
+    # correct treatment of arbitrary time limits (DM-Control-style tasks).
+    # True = treat the time limit as termination (classic episodic trick):
+    # anchors the value scale at episode ends, which grounds Q thousands of
+    # steps faster on short-horizon tasks like Pendulum (measured: PEB's Q
+    # sinks unanchored toward r/(1-gamma) while the policy stays degenerate
+    # at the smoke test's step budget; the episodic objective solves it).
+    timeout_terminal: bool = False
 
 
 @dataclass

@@ -11,14 +11,23 @@ initial priority from the actor's own step-wise Q values:
     p0      = eta * max|delta| + (1-eta) * mean|delta|
 
 At episode end the window is flushed (right-padded to fixed length by
-repeating the last observation with zero reward and done=0) and reset.
-Termination is carried solely by the REAL last step's done flag: after a
-terminal step the n-step fold's alive-mask is already 0, while after a
-time-limit truncation (done=0) the fold keeps bootstrapping through the
-repeated final observation — i.e. from the target net's value of the last
-real state — instead of wrongly treating truncation as termination. Padded
-steps are excluded from the critic loss and the priority reduction via
-``Sequence.length`` (carried through replay into the learner's mask).
+repeating the last observation with zero reward and zero action) and
+reset. Pad semantics depend on WHY the episode ended:
+
+* **Termination** (env terminal, or time limit with
+  ``replay.timeout_terminal``): pads are the ABSORBING state — done=1,
+  r=0, and they COUNT as real steps (``length`` = stored length), so the
+  learner trains Q(absorbing, ·) → 0. This is the standard absorbing-state
+  formalism and it is load-bearing: supervising the zero anchor directly
+  grounds the critic's value scale orders of magnitude faster than letting
+  it propagate only through bootstrapped folds (measured on the Pendulum
+  smoke config — without it the critic sinks toward r/(1−γ) for thousands
+  of steps while the policy stays degenerate).
+* **Truncation** (time limit, PEB): pads keep done=0 and are EXCLUDED
+  from the critic/actor losses and the priority reduction via
+  ``Sequence.length``; the n-step fold bootstraps through the repeated
+  final observation — i.e. from the target net's value of the last real
+  state — instead of wrongly treating truncation as termination.
 """
 
 from __future__ import annotations
@@ -82,28 +91,31 @@ class SequenceCutter:
             self.q = self.q[k:]
             self.states = self.states[k:]
         if done:
-            seqs = self.flush()
+            seqs = self.flush(terminated=True)
             out.extend(seqs)
         return out
 
-    def flush(self) -> List[Sequence]:
-        """Emit the remaining window (padded) at episode end, then reset."""
+    def flush(self, terminated: bool = False) -> List[Sequence]:
+        """Emit the remaining window (padded) at episode end, then reset.
+
+        ``terminated`` selects absorbing-state pads (trained, done=1) vs
+        truncation pads (masked, done=0) — module docstring."""
         out: List[Sequence] = []
         # only emit if there is at least one step beyond burn-in to train on
         if len(self.obs) > self.burn_in + 1:
-            out.append(self._emit(len(self.obs)))
+            out.append(self._emit(len(self.obs), absorbing=terminated))
         self._reset_window()
         return out
 
-    def _emit(self, length: int) -> Sequence:
+    def _emit(self, length: int, absorbing: bool = False) -> Sequence:
         L = self.stored_len
         obs_arr = np.zeros((L,) + self.obs[0].shape, dtype=self.obs[0].dtype)
         act_arr = np.zeros((L, self.act[0].shape[0]), dtype=np.float32)
         rew_arr = np.zeros(L, dtype=np.float32)
-        # pad region: done=0 (termination lives in the real flags; with the
-        # repeated-obs pad this bootstraps truncated episodes from the last
-        # real state's target value — see module docstring)
-        done_arr = np.zeros(L, dtype=np.float32)
+        # pad semantics (module docstring): terminated -> absorbing state
+        # (done=1, counts as real -> the learner trains Q(absorb)->0);
+        # truncated -> done=0, masked, bootstraps through the repeated obs
+        done_arr = np.full(L, 1.0 if absorbing else 0.0, dtype=np.float32)
         n = min(length, L)
         obs_arr[:n] = np.stack(self.obs[:n])
         act_arr[:n] = np.stack(self.act[:n])
@@ -113,7 +125,8 @@ class SequenceCutter:
             obs_arr[n:] = obs_arr[n - 1]  # repeat last obs into the pad
         prio = self._initial_priority(rew_arr, done_arr, np.asarray(self.q[:n], dtype=np.float32), n)
         return Sequence(obs_arr, act_arr, rew_arr, done_arr,
-                        self.states[0].copy(), float(prio), n)
+                        self.states[0].copy(), float(prio),
+                        L if absorbing else n)
 
     def _initial_priority(self, rew, done, q, n_real) -> float:
         """n-step TD error against the actor's own step-wise Q estimates.

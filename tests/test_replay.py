@@ -111,6 +111,9 @@ def test_cutter_emits_at_stride():
 
 
 def test_cutter_flush_on_done_pads():
+    """TERMINATED flush: pads are the absorbing state — done=1, trained
+    (length = stored_len so the learner does NOT mask them; the Q->0 anchor
+    at absorbing states grounds the critic's value scale)."""
     c = _mk_cutter()
     emitted = []
     for t in range(7):
@@ -119,12 +122,31 @@ def test_cutter_flush_on_done_pads():
                           1.0, done, 0.0, np.zeros((4, 4), np.float32))
     assert len(emitted) == 1  # flushed partial
     s = emitted[0]
-    assert s.length == 7
+    assert s.length == 10     # absorbing pads count as real steps
     assert s.obs.shape[0] == 10
-    assert s.done[6] == 1.0  # the REAL terminal flag
-    assert s.done[7] == 0.0  # pad done=0 (masked out via s.length instead)
+    assert s.done[6] == 1.0   # the REAL terminal flag
+    assert s.done[7] == 1.0 and s.done[9] == 1.0  # absorbing pads
+    assert (s.reward[7:] == 0).all()
     assert s.obs[8, 0] == 6.0  # pad repeats last obs
     # window reset after done
+    assert len(c.obs) == 0
+
+
+def test_cutter_flush_on_truncation_masks_pads():
+    """TRUNCATED flush (PEB): pads keep done=0 and are excluded from the
+    losses via length < stored_len; the learner bootstraps through the
+    repeated final observation."""
+    c = _mk_cutter()
+    emitted = []
+    for t in range(7):
+        emitted += c.push(np.full(3, t, dtype=np.float32), np.zeros(2, np.float32),
+                          1.0, False, 0.0, np.zeros((4, 4), np.float32))
+    emitted += c.flush(terminated=False)
+    assert len(emitted) == 1
+    s = emitted[0]
+    assert s.length == 7      # real steps only -> learner masks the pads
+    assert (s.done == 0).all()
+    assert s.obs[8, 0] == 6.0
     assert len(c.obs) == 0
 
 

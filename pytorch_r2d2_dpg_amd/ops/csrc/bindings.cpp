@@ -194,15 +194,27 @@ torch::Tensor pack_fragment_major(const torch::Tensor& w) {
 // are held by the map, so the caching allocator never recycles them.
 // Keyed by data_ptr: weights are long-lived parameters whose storage is
 // stable; a version bump also clears entries whose shape changed.
-struct PackEntry { uint64_t version; torch::Tensor packed; };
+// Entries PIN the source tensor (`src`): while cached, its storage cannot
+// be freed, so a later allocation can never reuse the data_ptr key — the
+// classic stale-hit hazard for tensors created/dropped outside the learner
+// (unit tests). A hit additionally requires BOTH the learner's explicit
+// version (covers the K4 Polyak raw-pointer writes torch can't see) and
+// the tensor's autograd version counter (covers in-place torch mutation
+// without a bump) to match.
+struct PackEntry {
+    uint64_t gver;
+    uint32_t tver;
+    torch::Tensor src;
+    torch::Tensor packed;
+};
 uint64_t g_weight_version = 0;
 std::unordered_map<const void*, PackEntry> g_pack_cache;    // w -> pack(w)
 std::unordered_map<const void*, PackEntry> g_pack_cache_T;  // w -> pack(w^T)
 
 void bump_weight_version() {
     ++g_weight_version;
-    if (g_pack_cache.size() > 512) g_pack_cache.clear();      // leak guard
-    if (g_pack_cache_T.size() > 512) g_pack_cache_T.clear();
+    if (g_pack_cache.size() > 64) g_pack_cache.clear();       // bound pinning
+    if (g_pack_cache_T.size() > 64) g_pack_cache_T.clear();
 }
 
 bool stream_capturing() {
@@ -220,13 +232,15 @@ torch::Tensor pack_cached(const torch::Tensor& w, bool transpose) {
                          : pack_fragment_major(w);
     auto& cache = transpose ? g_pack_cache_T : g_pack_cache;
     const void* key = w.data_ptr();
+    const uint32_t tver = w.unsafeGetTensorImpl()->version_counter().current_version();
     auto it = cache.find(key);
-    if (it != cache.end() && it->second.version == g_weight_version &&
+    if (it != cache.end() && it->second.gver == g_weight_version &&
+        it->second.tver == tver && it->second.src.data_ptr() == key &&
         it->second.packed.numel() == w.numel())
         return it->second.packed;
     auto packed = transpose ? pack_fragment_major(w.t().contiguous())
                             : pack_fragment_major(w);
-    cache[key] = {g_weight_version, packed};
+    cache[key] = {g_weight_version, tver, w, packed};
     return packed;
 }
 

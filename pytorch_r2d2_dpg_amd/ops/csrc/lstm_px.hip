@@ -221,11 +221,30 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     }
 }
 
-// All px launchers take a CU budget (<= 224) from the caller's in-flight
-// accounting (bindings.cpp): every block of a px grid must be co-resident
-// with all its peers, so the sum of concurrently-running px blocks must
-// never exceed the chip. Success returns the BLOCK COUNT (positive) so the
-// caller can commit it against the budget; failure returns -1.
+// All px launchers take a CU budget (<= 224 CU-equivalents) from the
+// caller's in-flight accounting (bindings.cpp): every block of a px grid
+// must be co-resident with all its peers, so the CU capacity consumed by
+// concurrently-running px grids must never exceed the chip. Capacity is
+// counted in CU-EQUIVALENTS: a CU fits several px blocks (LDS- and
+// wave-slot-limited — e.g. 5 fwd blocks at 32 KB LDS each, 2 bwd blocks at
+// 64 KB), so cu_eq = ceil(blocks / slots_per_cu). r1's raw-block rule
+// (R2D2_PX_RAWBLOCKS=1 restores it) forced pairs down to CG=4 and starved
+// concurrent launches; the slot model lets the whole step run at CG=8.
+// Success returns the grid's CU-equivalents (positive) for the caller to
+// commit; failure returns -1.
+static inline int px_cu_eq(int blocks, int waves_pb, long lds_bytes) {
+    static const bool raw = []() {
+        const char* e = getenv("R2D2_PX_RAWBLOCKS");
+        return e && e[0] && e[0] != '0';
+    }();
+    if (raw) return blocks;
+    int slots = (int)(160L * 1024 / (lds_bytes > 0 ? lds_bytes : 1));
+    int wslots = 32 / (waves_pb > 0 ? waves_pb : 1);
+    if (wslots < slots) slots = wslots;
+    if (slots > 8) slots = 8;
+    if (slots < 1) slots = 1;
+    return (blocks + slots - 1) / slots;
+}
 extern "C" int launch_lstm_fwd_px_bf16(
     const void* gates_x, const void* h0, const void* c0, const void* wh_packed,
     void* h_seq, void* c_final, void* gates_act, float* c_seq,
@@ -235,13 +254,15 @@ extern "C" int launch_lstm_fwd_px_bf16(
     // largest CG with co-residency margin and >= 2 fragments per gate
     if (H & (H - 1)) return -1;              // sweep indexing needs pow2 H
     if (budget > 224) budget = 224;
-    int CG = 0;
+    int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;          // wave owns 16 cols
         int w = (H / c) / 16;                // waves per block
         if (w < 2 || w > 16) continue;
-        if (RT * c > budget) continue;       // all blocks must be co-resident
+        int ce = px_cu_eq(RT * c, w, 2L * 16 * H * 2);
+        if (ce > budget) continue;           // co-residency budget
         CG = c;
+        cu_eq = ce;
         break;
     }
     if (CG == 0) return -1;
@@ -274,7 +295,7 @@ extern "C" int launch_lstm_fwd_px_bf16(
             return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return RT * CG;
+    return cu_eq;
 }
 
 // ---------------------------------------------------------------- backward
@@ -553,7 +574,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
     if (H & (H - 1)) return -1;
     if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;  // keep 2 blocks/CU
     if (budget > 224) budget = 224;
-    int CG = 0;
+    int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
@@ -562,8 +583,10 @@ extern "C" int launch_lstm_bwd_px_bf16(
                                    // swizzle permutes 16B blocks within a
                                    // 128B window, so raw byte-range copies
                                    // are closed only for >=128B regions
-        if (RT * c > budget) continue;
+        int ce = px_cu_eq(RT * c, w, 16L * 4 * H * 2);
+        if (ce > budget) continue;
         CG = c;
+        cu_eq = ce;
         break;
     }
     if (CG == 0) return -1;
@@ -582,7 +605,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return RT * CG;
+    return cu_eq;
 }
 
 extern "C" int launch_lstm_bwd_px2_bf16(
@@ -597,14 +620,16 @@ extern "C" int launch_lstm_bwd_px2_bf16(
     if (H & (H - 1)) return -1;
     if ((size_t)16 * 4 * H * 2 > 128 * 1024) return -1;
     if (budget > 224) budget = 224;
-    int CG = 0;
+    int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
         if (H / c < 64) continue;        // slab swizzle closure (see solo)
-        if (RT * c * 2 > budget) continue;  // BOTH halves must co-reside
+        int ce = px_cu_eq(RT * c * 2, w, 16L * 4 * H * 2);
+        if (ce > budget) continue;       // BOTH halves must co-reside
         CG = c;
+        cu_eq = ce;
         break;
     }
     if (CG == 0) return -1;
@@ -628,7 +653,7 @@ extern "C" int launch_lstm_bwd_px2_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return RT * CG * 2;
+    return cu_eq;
 }
 
 // ------------------------------------------------- paired (z=2) forward
@@ -826,13 +851,15 @@ extern "C" int launch_lstm_fwd_px2_bf16(
     int RT = (B + 15) / 16;
     if (H & (H - 1)) return -1;
     if (budget > 224) budget = 224;
-    int CG = 0;
+    int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
-        if (RT * c * 2 > budget) continue;  // BOTH halves must co-reside
+        int ce = px_cu_eq(RT * c * 2, w, 2L * 16 * H * 2);
+        if (ce > budget) continue;       // BOTH halves must co-reside
         CG = c;
+        cu_eq = ce;
         break;
     }
     if (CG == 0) return -1;
@@ -854,5 +881,5 @@ extern "C" int launch_lstm_fwd_px2_bf16(
         default: return -1;
     }
     LSTM_CHECK_LAUNCH();
-    return RT * CG * 2;
+    return cu_eq;
 }

@@ -11,9 +11,12 @@
 // delta vs online q; then wave-reduce max|d| and sum|d| -> priority.
 // Optional value-rescale: y = h(folded + g^n * mask * h_inv(q_boot)) applied
 // HERE so the priority/TD math matches the learner exactly.
-// valid (nullable, [B]): count of REAL (non-pad) training-window steps per
-// sequence — pad steps at the tail of flushed episodes contribute neither
-// td_abs nor priority mass (the learner masks them out of the loss too).
+// valid (nullable, [B]): count of REAL (non-pad) steps per sequence counted
+// from the window start, UNCLAMPED (may exceed T into the n-step tail).
+// Steps >= min(valid, T) contribute neither td_abs nor priority mass (the
+// learner masks them out of the loss too), and the reward fold treats
+// indices >= valid as r=0/done=0 — stored pad bytes can never leak into a
+// real step's target.
 extern "C" __global__ void __launch_bounds__(64)
 td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
                   const float* __restrict__ dones,    // [B, t_ext]
@@ -34,10 +37,11 @@ td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
     const float* qo_row = q_online + (long)b * T;
     float gamma_n = 1.0f;
     for (int k = 0; k < n_step; ++k) gamma_n *= gamma;
+    int v_ext = t_ext;  // real extent of the reward/done rows
     int vT = T;
     if (valid) {
-        vT = valid[b];
-        vT = vT < 1 ? 1 : (vT > T ? T : vT);
+        v_ext = valid[b] < 1 ? 1 : valid[b];
+        vT = v_ext > T ? T : v_ext;
     }
 
     float max_d = 0.0f, sum_d = 0.0f;
@@ -46,8 +50,10 @@ td_targets_kernel(const float* __restrict__ rewards,  // [B, t_ext]
         for (int k = 0; k < n_step; ++k) {
             int idx = t + k;
             if (idx >= t_ext) { alive = 0.0f; break; }
-            acc += g * alive * r_row[idx];
-            alive *= 1.0f - d_row[idx];
+            if (idx < v_ext) {  // beyond the real extent: r=0, done=0
+                acc += g * alive * r_row[idx];
+                alive *= 1.0f - d_row[idx];
+            }
             g *= gamma;
         }
         float qb = qb_row[t];

@@ -246,3 +246,32 @@ def test_orchestrated_two_rank_cpu(tmp_path):
     step_lines = [l for l in out.splitlines() if "[step " in l]
     assert step_lines, out[-3000:]
     assert all("actor_respawns=0" in l for l in step_lines), step_lines
+
+
+def test_bench_two_rank_gloo_cpu():
+    """bench.py under torchrun world 2 on gloo/CPU — the exact entrypoint
+    and launch shape the round-end SCALE driver uses:
+    DistContext.from_env, fused flat-bucket all-reduces, max-over-ranks
+    timing, one JSON line from rank 0."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "2", "--standalone", "--local-addr", "127.0.0.1",
+        os.path.join(root, "bench.py"),
+        "--gpus", "2", "--steps", "3", "--warmup", "1",
+        "--batch", "8", "--seq-len", "12", "--burn-in", "4",
+        "--hidden", "64", "--dtype", "fp32", "--replay-capacity", "64",
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=420, cwd=root)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert line, out[-3000:]
+    d = json.loads(line[-1])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0 and d["ms_per_step"] > 0

@@ -499,3 +499,30 @@ def test_lstm_pair_halfgrad_matches_singles():
                             (c1[0], c1[3], c1[4], c1[5]),
                             (c2[0], c2[3], c2[4], c2[5])):
         assert p1.grad is not None and torch.equal(p1.grad, p2.grad), name
+
+
+@pytest.mark.gpu
+def test_td_targets_valid_mask_matches_cpu_oracle():
+    """K3 with a valid-count input vs the CPU reference path: pad steps get
+    td_abs = 0 and the priority reduction uses only the valid prefix."""
+    from pytorch_r2d2_dpg_amd.ops.td import td_targets_and_priorities
+
+    B, T, n = 16, 20, 5
+    t_ext = T + n
+    g = torch.Generator().manual_seed(5)
+    rew = torch.randn(B, t_ext, generator=g)
+    done = (torch.rand(B, t_ext, generator=g) < 0.1).float()
+    qb = torch.randn(B, T, generator=g)
+    qo = torch.randn(B, T, generator=g)
+    valid = torch.randint(1, T + 1, (B,), generator=g, dtype=torch.int32)
+    y_c, td_c, p_c = td_targets_and_priorities(rew, done, qb, qo, 0.997, n,
+                                               0.9, valid_n=valid)
+    y_g, td_g, p_g = td_targets_and_priorities(
+        rew.cuda(), done.cuda(), qb.cuda(), qo.cuda(), 0.997, n, 0.9,
+        valid_n=valid.cuda())
+    assert torch.allclose(y_g.cpu(), y_c, atol=2e-5, rtol=1e-5)
+    assert torch.allclose(td_g.cpu(), td_c, atol=2e-5, rtol=1e-5)
+    assert torch.allclose(p_g.cpu(), p_c, atol=2e-5, rtol=1e-5)
+    # pad region carries no priority mass
+    for b in range(B):
+        assert (td_g[b, int(valid[b]):].cpu() == 0).all()

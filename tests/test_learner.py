@@ -169,3 +169,62 @@ def test_manual_decomposition_matches_module_forward():
     assert torch.equal(hs1, r)
     r.square().mean().backward()
     assert torch.allclose(x1.grad, x2.grad)
+
+
+def test_truncation_pad_steps_cannot_influence_update():
+    """Masked-loss contract: for a TRUNCATED (PEB) sequence, nothing in the
+    pad region may influence the parameter update — scrambling pad rewards
+    and actions must yield the identical step (the repeated pad OBS is the
+    bootstrap source and is kept identical here)."""
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    torch.manual_seed(0)
+    cfg = Config()
+    cfg.env.name = "synthetic"
+    cfg.model.hidden_size = 32
+    cfg.model.trunk_sizes = (32,)
+    cfg.replay.seq_len = 12
+    cfg.replay.burn_in = 4
+    cfg.replay.n_step = 3
+    cfg.learner.batch_size = 6
+    spec = make_env(cfg.env.name).spec
+    L = cfg.replay.seq_len + cfg.replay.n_step
+    B, H = cfg.learner.batch_size, cfg.model.hidden_size
+    g = torch.Generator().manual_seed(7)
+    length = torch.tensor([L, 9, L, 11, 7, L], dtype=torch.int32)
+    base = {
+        "obs": torch.randn(B, L, *spec.obs_shape, generator=g),
+        "action": torch.randn(B, L, spec.action_dim, generator=g),
+        "reward": torch.rand(B, L, generator=g),
+        "done": torch.zeros(B, L),
+        "state": torch.randn(B, 4, H, generator=g) * 0.05,
+        "weight": torch.ones(B),
+        "idx": None,
+        "length": length,
+    }
+    # repeat the last real obs into the pad (as the cutter does) and zero
+    # pad rewards so the two runs share the bootstrap inputs
+    for b in range(B):
+        n = int(length[b])
+        if n < L:
+            base["obs"][b, n:] = base["obs"][b, n - 1]
+            base["reward"][b, n:] = 0.0
+
+    results = []
+    for scramble in (False, True):
+        batch = {k: (v.clone() if torch.is_tensor(v) else v) for k, v in base.items()}
+        if scramble:
+            gs = torch.Generator().manual_seed(99)
+            for b in range(B):
+                n = int(length[b])
+                if n < L:
+                    batch["reward"][b, n:] = torch.randn(L - n, generator=gs) * 50
+                    batch["action"][b, n:] = torch.randn(L - n, spec.action_dim,
+                                                         generator=gs) * 3
+        torch.manual_seed(3)
+        lrn = Learner(cfg, spec, device="cpu")
+        lrn.train_step(batch)
+        results.append(torch.cat([p.detach().reshape(-1)
+                                  for p in list(lrn.actor.parameters())
+                                  + list(lrn.critic.parameters())]))
+    assert torch.equal(results[0], results[1]), \
+        "pad rewards/actions leaked into the update"

@@ -202,7 +202,10 @@ class Learner:
         length = batch.get("length")
         if length is not None:
             T = self.train_len
-            valid = (length.to(torch.int32) - self.burn_in).clamp(1, T)
+            # valid = real steps from the window start, UNCLAMPED into the
+            # n-step tail (K3 masks reward reads beyond it); the loss mask
+            # clamps to the training window
+            valid = (length.to(torch.int32) - self.burn_in).clamp(1, T + self.n_step)
             vmask = (torch.arange(T, device=obs.device).unsqueeze(0)
                      < valid.to(torch.int64).unsqueeze(1)).to(torch.float32)
         else:

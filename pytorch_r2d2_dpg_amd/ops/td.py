@@ -34,8 +34,13 @@ def td_targets_and_priorities(
     eta: float,
     rescale: bool = False,        # R2D2 value rescaling h(.) (flag, default off)
     rescale_eps: float = 1e-3,
-    valid_n: torch.Tensor = None,  # [B] int32: real (non-pad) training steps;
-                                   # pad steps get td_abs=0 and no priority mass
+    valid_n: torch.Tensor = None,  # [B] int32: real (non-pad) steps from the
+                                   # window start, UNCLAMPED (may reach into
+                                   # the n-step tail). Steps >= min(valid, T)
+                                   # get td_abs=0 and no priority mass; the
+                                   # reward fold treats indices >= valid as
+                                   # r=0/done=0 so stored pad bytes never
+                                   # leak into a real step's target.
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Returns (targets y [B,T], |delta| [B,T], priorities [B])."""
     T = q_online.shape[1]
@@ -46,6 +51,13 @@ def td_targets_and_priorities(
             rescale, rescale_eps,
             valid_n.contiguous() if valid_n is not None else None,
         )
+    if valid_n is not None:
+        # mirror the kernel's reward-extent guard
+        t_ext = rewards.shape[1]
+        ext_mask = (torch.arange(t_ext, device=rewards.device).unsqueeze(0)
+                    < valid_n.to(torch.long).clamp(min=1).unsqueeze(1)).to(rewards.dtype)
+        rewards = rewards * ext_mask
+        dones = dones * ext_mask
     qb = value_rescale_inv(bootstrap_q, rescale_eps) if rescale else bootstrap_q
     folded, mask = fold_nstep_ref(rewards, dones, gamma, n_step, T)
     y = folded + (gamma ** n_step) * mask * qb

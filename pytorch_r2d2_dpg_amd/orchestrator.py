@@ -149,6 +149,7 @@ def run_training(cfg: Config, device: Optional[str] = None, resume: Optional[str
                     eval_return=float(np.mean([r for _, r in pool.eval_returns[-10:]]))
                     if pool.eval_returns else 0.0,
                     actor_respawns=pool.respawn_count,
+                    ring_drops=pool.aggregate_drops(),
                     px_err=learner.px_err_total,
                 )
                 t_last, trans_acc, env_steps_last = now, 0, env_steps

@@ -253,3 +253,7 @@ class ActorPool:
 
     def aggregate_env_steps(self) -> int:
         return int(sum(self.heartbeat))
+
+    def aggregate_drops(self) -> int:
+        """Total sequences dropped to ring backpressure (drop-newest)."""
+        return int(sum(r.dropped.value for r in self.rings))

@@ -295,7 +295,7 @@ lstm_seq_fwd(torch::Tensor gates_x, torch::Tensor h0, torch::Tensor c0,
                 (int)T, (int)B, (int)H, stash ? 1 : 0, bud, cur_stream());
         };
         rc = try_px(budget);
-        if (rc < 0 && budget < 224)  // contended: serialize behind in-flight px
+        if (rc == -2)  // would over-subscribe: serialize behind in-flight px
             rc = try_px(px_budget_drain(cur_stream()));
         if (rc > 0) {
             px_budget_commit(rc, cur_stream());
@@ -368,7 +368,7 @@ lstm_seq_fwd_pair(torch::Tensor gx0, torch::Tensor h00, torch::Tensor c00,
             cur_stream());
     };
     int rc = try_px(budget);
-    if (rc < 0 && budget < 224)
+    if (rc == -2)
         rc = try_px(px_budget_drain(cur_stream()));
     TORCH_CHECK(rc > 0, "pair geometry unsupported (fall back to two calls)");
     px_budget_commit(rc, cur_stream());
@@ -420,7 +420,7 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
                 (int)T, (int)B, (int)H, bud, cur_stream());
         };
         rc = try_px(budget);
-        if (rc < 0 && budget < 224)
+        if (rc == -2)
             rc = try_px(px_budget_drain(cur_stream()));
         if (rc > 0) {
             px_budget_commit(rc, cur_stream());
@@ -498,7 +498,7 @@ lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
             cur_stream());
     };
     int rc = try_px(budget);
-    if (rc < 0 && budget < 224)
+    if (rc == -2)
         rc = try_px(px_budget_drain(cur_stream()));
     TORCH_CHECK(rc > 0, "bwd pair geometry unsupported (fall back to two calls)");
     px_budget_commit(rc, cur_stream());

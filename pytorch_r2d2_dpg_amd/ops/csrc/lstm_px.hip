@@ -221,23 +221,16 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     }
 }
 
-// All px launchers take a CU budget (<= 224 CU-equivalents) from the
-// caller's in-flight accounting (bindings.cpp): every block of a px grid
-// must be co-resident with all its peers, so the CU capacity consumed by
-// concurrently-running px grids must never exceed the chip. Capacity is
-// counted in CU-EQUIVALENTS: a CU fits several px blocks (LDS- and
-// wave-slot-limited — e.g. 5 fwd blocks at 32 KB LDS each, 2 bwd blocks at
-// 64 KB), so cu_eq = ceil(blocks / slots_per_cu). r1's raw-block rule
-// (R2D2_PX_RAWBLOCKS=1 restores it) forced pairs down to CG=4 and starved
-// concurrent launches; the slot model lets the whole step run at CG=8.
-// Success returns the grid's CU-equivalents (positive) for the caller to
-// commit; failure returns -1.
+// CG SELECTION is budget-independent (the r1 rule, RT*CG(*2) <= 224 —
+// measured fastest: budget-driven CG reduction cost the flagship 17%).
+// The budget is a SAFETY NET only: capacity is counted in CU-EQUIVALENTS
+// (a CU fits several px blocks, LDS- and wave-slot-limited — e.g. 5 fwd
+// blocks at 32 KB LDS each, 2 bwd blocks at 64 KB; cu_eq =
+// ceil(blocks/slots)), and a launch whose cu_eq exceeds the REMAINING
+// budget returns -2 so the caller serializes it behind the in-flight px
+// grids instead of risking a co-residency livelock. Success returns the
+// grid's CU-equivalents (positive) for the caller to commit.
 static inline int px_cu_eq(int blocks, int waves_pb, long lds_bytes) {
-    static const bool raw = []() {
-        const char* e = getenv("R2D2_PX_RAWBLOCKS");
-        return e && e[0] && e[0] != '0';
-    }();
-    if (raw) return blocks;
     int slots = (int)(160L * 1024 / (lds_bytes > 0 ? lds_bytes : 1));
     int wslots = 32 / (waves_pb > 0 ? waves_pb : 1);
     if (wslots < slots) slots = wslots;
@@ -259,13 +252,13 @@ extern "C" int launch_lstm_fwd_px_bf16(
         if (H % (16 * c)) continue;          // wave owns 16 cols
         int w = (H / c) / 16;                // waves per block
         if (w < 2 || w > 16) continue;
-        int ce = px_cu_eq(RT * c, w, 2L * 16 * H * 2);
-        if (ce > budget) continue;           // co-residency budget
+        if (RT * c > 224) continue;          // all blocks must be co-resident
         CG = c;
-        cu_eq = ce;
+        cu_eq = px_cu_eq(RT * c, w, 2L * 16 * H * 2);
         break;
     }
     if (CG == 0) return -1;
+    if (cu_eq > budget) return -2;           // caller: drain in-flight, retry
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG);
     size_t lds = (size_t)2 * 16 * H * 2;
@@ -583,13 +576,13 @@ extern "C" int launch_lstm_bwd_px_bf16(
                                    // swizzle permutes 16B blocks within a
                                    // 128B window, so raw byte-range copies
                                    // are closed only for >=128B regions
-        int ce = px_cu_eq(RT * c, w, 16L * 4 * H * 2);
-        if (ce > budget) continue;
+        if (RT * c > 224) continue;
         CG = c;
-        cu_eq = ce;
+        cu_eq = px_cu_eq(RT * c, w, 16L * 4 * H * 2);
         break;
     }
     if (CG == 0) return -1;
+    if (cu_eq > budget) return -2;           // caller: drain in-flight, retry
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG);
     size_t lds = (size_t)16 * 4 * H * 2;
@@ -626,13 +619,13 @@ extern "C" int launch_lstm_bwd_px2_bf16(
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
         if (H / c < 64) continue;        // slab swizzle closure (see solo)
-        int ce = px_cu_eq(RT * c * 2, w, 16L * 4 * H * 2);
-        if (ce > budget) continue;       // BOTH halves must co-reside
+        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
         CG = c;
-        cu_eq = ce;
+        cu_eq = px_cu_eq(RT * c * 2, w, 16L * 4 * H * 2);
         break;
     }
     if (CG == 0) return -1;
+    if (cu_eq > budget) return -2;       // caller: drain in-flight, retry
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG, 2);
     size_t lds = (size_t)16 * 4 * H * 2;
@@ -856,13 +849,13 @@ extern "C" int launch_lstm_fwd_px2_bf16(
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
-        int ce = px_cu_eq(RT * c * 2, w, 2L * 16 * H * 2);
-        if (ce > budget) continue;       // BOTH halves must co-reside
+        if (RT * c * 2 > 224) continue;  // BOTH halves must co-reside
         CG = c;
-        cu_eq = ce;
+        cu_eq = px_cu_eq(RT * c * 2, w, 2L * 16 * H * 2);
         break;
     }
     if (CG == 0) return -1;
+    if (cu_eq > budget) return -2;       // caller: drain in-flight, retry
     int waves = (H / CG) / 16;
     dim3 grid(RT, CG, 2);
     size_t lds = (size_t)2 * 16 * H * 2;

@@ -103,6 +103,7 @@ void set_px_serialize(bool on) { g_px_serialize = on; }
 // CONCURRENT over-subscribed px livelocks until the spin bound). GIL
 // serializes all access. R2D2_PX_NO_BUDGET=1 restores the fixed 224 limit.
 struct PxInflight { hipEvent_t ev; int blocks; hipStream_t st; };
+long g_px_launches = 0, g_px_drains = 0, g_px_reserve_min = 224;
 std::vector<PxInflight> g_px_inflight;       // current device only (per-proc 1 GPU)
 std::vector<hipEvent_t> g_px_event_pool;
 bool g_px_budget_enabled = []() {
@@ -118,6 +119,7 @@ bool g_px_budget_enabled = []() {
 int px_budget_reserve(hipStream_t cur) {
     if (!g_px_budget_enabled) return 224;
     int used = 0;
+
     for (size_t i = 0; i < g_px_inflight.size();) {
         if (hipEventQuery(g_px_inflight[i].ev) == hipSuccess) {
             g_px_event_pool.push_back(g_px_inflight[i].ev);
@@ -135,6 +137,7 @@ int px_budget_reserve(hipStream_t cur) {
 // make `st` wait for every OTHER-stream in-flight px grid, then return the
 // full budget (same-stream grids are already ordered before the launch)
 int px_budget_drain(hipStream_t st) {
+    ++g_px_drains;
     for (auto& f : g_px_inflight)
         if (f.st != st)
             (void)hipStreamWaitEvent(st, f.ev, 0);
@@ -142,6 +145,7 @@ int px_budget_drain(hipStream_t st) {
 }
 
 void px_budget_commit(int blocks, hipStream_t st) {
+    ++g_px_launches;
     if (!g_px_budget_enabled) return;
     hipEvent_t ev;
     if (!g_px_event_pool.empty()) {
@@ -162,6 +166,12 @@ torch::Tensor& px_err_word(const torch::Device& dev) {
     if (it == words.end())
         it = words.emplace(idx, torch::zeros({1}, torch::dtype(torch::kInt32).device(dev))).first;
     return it->second;
+}
+
+std::tuple<int64_t, int64_t, int64_t> px_budget_stats() {
+    auto r = std::make_tuple((int64_t)g_px_launches, (int64_t)g_px_drains,
+                             (int64_t)g_px_inflight.size());
+    return r;
 }
 
 // Read-and-clear the column-split spin-timeout error word (0 = healthy).
@@ -646,6 +656,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "two independent LSTM backwards in one launch (per-set T)");
     m.def("set_px_serialize", &set_px_serialize,
           "serialize column-split kernels device-wide (off by default)");
+    m.def("px_budget_stats", &px_budget_stats,
+          "(px launches, budget drains, inflight entries)");
     m.def("bump_weight_version", &bump_weight_version,
           "invalidate the fragment-major weight-pack cache (call after every "
           "optimizer/Polyak write)");

@@ -417,11 +417,13 @@ class Learner:
         w_head = self.critic.head.weight          # [1, H]
         b_head = self.critic.head.bias
 
-        # ---- side stream: actor chain FORWARD + the raw critic GATES only.
-        # The critic unroll itself (QPI) is PAIRED with the target-critic
-        # unroll on main below — the two become ready within ~0.2 ms of
-        # each other, and one grid.z=2 launch replaces two solo px grids
-        # (docs/KERNELS.md; r1 roadmap lever 1b).
+        # ---- side stream: actor chain FORWARD only + the q_pi loss seed.
+        # The critic unroll here is a RAW stash forward (its weight grads
+        # are discarded by design — pre-update-critic DDPG), so only dmu is
+        # needed from its backward, which the paired K2 below provides. The
+        # loss seed d(actor_loss)/d(hs_q) is ANALYTIC (-m/Σm ⊗ w_head): no
+        # autograd graph touches the critic head here, which keeps the head
+        # params' AccumulateGrad nodes on the main stream (docs/STREAMS.md).
         with torch.cuda.stream(s1):
             s1.wait_event(self._ev_ba)
             la = self.actor.lstm
@@ -436,14 +438,23 @@ class Learner:
                 B_ = x_q.shape[1]
                 gx_q = torch.addmm(b_c, x_q.reshape(T * B_, -1),
                                    wx_c.t()).view(T, -1, 4 * H).contiguous()
+                hs_q, _, ga_q, cs_q = ext.lstm_seq_fwd(
+                    gx_q, h_c2.detach().contiguous(), c_c2d, wh_c.contiguous())
+                q_pi = (hs_q.reshape(-1, H) @ w_head.t() + b_head).view(T, B_)
+                actor_loss = self._actor_loss(q_pi, vmask)
+                if vmask is None:
+                    dhs_q = (w_head.mul(-1.0 / (T * B_)).to(hs_q.dtype)
+                             .view(1, 1, H).expand(T, B_, H).contiguous())
+                else:
+                    m_tb = vmask.t()
+                    dhs_q = ((m_tb / -m_tb.sum()).unsqueeze(-1).to(hs_q.dtype)
+                             * w_head.view(1, 1, H)).contiguous()
             self._ev_ab.record(s1)
 
         self._pe("burn_in", 1)
         self._pe("target", 0)
-        # ---- main: TA + CF (raw stash) paired fwd, then TC + QPI paired
-        # fwd, K3, critic loss
+        # ---- main: TA + CF (raw stash) paired fwd, TC, K3, critic loss
         ta = self.target_actor
-        tc = self.target_critic
         feat_cf = self.critic.trunk(obs_w)            # grad (trunk params)
         x_cf = torch.cat([feat_cf, action[bi:bi + T]], dim=-1)
         with torch.no_grad():
@@ -460,34 +471,8 @@ class Learner:
                 False, True)
             mu_t = (torch.tanh(ta.head(hs_ta.reshape(-1, H)))
                     .view(Tn, -1, ta.action_dim) * ta.action_scale)
-            # target-critic gates (raw, for the TC+QPI pair)
-            feat_tc = tc.trunk(obs[bi:bi + Tn])
-            x_tc = torch.cat([feat_tc, mu_t], dim=-1)
-            gx_tc = torch.addmm(tc.lstm.bias,
-                                x_tc.reshape(Tn * x_tc.shape[1], -1),
-                                tc.lstm.weight_x.t()).view(Tn, -1, 4 * H).contiguous()
-            # ---- TC + QPI in ONE paired launch (gx_q produced on side)
-            main.wait_event(self._ev_ab)
-            gx_q.record_stream(main)
-            hs_tc, _, _, _, hs_q, _, ga_q, cs_q = ext.lstm_seq_fwd_pair(
-                gx_tc, h_c2.contiguous(), c_c2.contiguous(),
-                tc.lstm.weight_h.contiguous(),
-                gx_q, h_c2.detach().contiguous(), c_c2d, wh_c.contiguous(),
-                False, True)
-            qt = (tc.head(hs_tc.reshape(-1, H)).view(Tn, -1))
+            qt, _ = self.target_critic(obs[bi:bi + Tn], mu_t, (h_c2, c_c2))
             q_boot = qt[self.n_step:self.n_step + T].transpose(0, 1).float()
-            # analytic actor-loss seed d(actor_loss)/d(hs_q) = -(m/Σm) ⊗
-            # w_head: no autograd graph touches the critic head on any
-            # stream (docs/STREAMS.md)
-            q_pi = (hs_q.reshape(-1, H) @ w_head.t() + b_head).view(T, B_)
-            actor_loss = self._actor_loss(q_pi, vmask)
-            if vmask is None:
-                dhs_q = (w_head.mul(-1.0 / (T * B_)).to(hs_q.dtype)
-                         .view(1, 1, H).expand(T, B_, H).contiguous())
-            else:
-                m_tb = vmask.t()
-                dhs_q = ((m_tb / -m_tb.sum()).unsqueeze(-1).to(hs_q.dtype)
-                         * w_head.view(1, 1, H)).contiguous()
         hs_cf_leaf = hs_cf.detach().requires_grad_(True)
         q_online = self.critic.head(hs_cf_leaf.reshape(-1, H)).view(T, -1)
         q_online_bt = q_online.transpose(0, 1).float()
@@ -505,7 +490,9 @@ class Learner:
             critic_loss, [hs_cf_leaf] + head_p)
 
         # ---- joint paired K2: both critic-LSTM backwards in one launch
-        # (all inputs main-produced now; ga_q/cs_q came from the TC+QPI pair)
+        main.wait_event(self._ev_ab)
+        for t_ in (dhs_q, ga_q, cs_q):
+            t_.record_stream(main)
         zero = torch.zeros_like(h_c2)
         try:
             dg_cf, _, _, dg_q, _, _ = ext.lstm_seq_bwd_pair(

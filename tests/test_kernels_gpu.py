@@ -526,3 +526,41 @@ def test_td_targets_valid_mask_matches_cpu_oracle():
     # pad region carries no priority mass
     for b in range(B):
         assert (td_g[b, int(valid[b]):].cpu() == 0).all()
+
+
+@pytest.mark.gpu
+def test_rccl_world1_collectives_smoke():
+    """RCCL (nccl backend) init + the exact collective path the DP learner
+    uses — fused flat-bucket all-reduce, broadcast_params, scalar
+    all-reduce — on a single-GPU world. Exercises the RCCL init plumbing
+    and stream interaction that gloo CPU tests cannot (VERDICT r1 item 1);
+    at world=1 the averaging all-reduce must be an exact identity."""
+    import os
+    import torch.distributed as dist
+    from pytorch_r2d2_dpg_amd.parallel.dist import DistContext
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    assert not dist.is_initialized()
+    dist.init_process_group("nccl", rank=0, world_size=1,
+                            device_id=torch.device("cuda", 0))
+    try:
+        ctx = DistContext(0, 1, True)
+        g = torch.Generator().manual_seed(0)
+        params = [torch.nn.Parameter(torch.randn(s, generator=g).cuda())
+                  for s in ((64, 32), (128,), (8, 8, 4))]
+        for p in params:
+            p.grad = torch.randn(p.shape, generator=g).cuda()
+        before = [p.grad.clone() for p in params]
+        ctx.allreduce_grads(params, key="smoke")
+        torch.cuda.synchronize()
+        for b, p in zip(before, params):
+            assert torch.allclose(b, p.grad), "world-1 all-reduce not identity"
+        w_before = [p.detach().clone() for p in params]
+        ctx.broadcast_params([p for p in params])
+        torch.cuda.synchronize()
+        for b, p in zip(w_before, params):
+            assert torch.equal(b, p.detach())
+        assert ctx.allreduce_scalar(3.5, op="sum") == 3.5
+    finally:
+        dist.destroy_process_group()

@@ -34,6 +34,26 @@ DEV_INLINE unsigned px_swz(unsigned byte_off, int row) {
     return byte_off ^ ((unsigned)(row & 7) << 4);
 }
 
+// XCD-aware (rt, cg) decode: the dispatcher round-robins blocks over the 8
+// XCDs by linear block id, so decoding cg = flat % CG puts every block of
+// one column group on the same XCD (CG=8: exactly one Wh slice per XCD
+// L2, resident across all T steps) or on CG/8-aligned subsets (CG=4/2) —
+// instead of every XCD streaming pieces of ALL slices and thrashing its
+// L2 with the gates/stash traffic. Placement-independent for correctness
+// (the exchange protocol only sees rt/cg); affinity is a speed lever only.
+// xcd_swz=0 restores the identity decode (A/B).
+template <int CG>
+DEV_INLINE void px_decode_block(int xcd_swz, int& rt, int& cg) {
+    if (xcd_swz) {
+        int flat = (int)(blockIdx.x + gridDim.x * blockIdx.y);
+        cg = flat % CG;
+        rt = flat / CG;
+    } else {
+        rt = blockIdx.x;
+        cg = blockIdx.y;
+    }
+}
+
 // One 16-col fragment per gate per wave; CG is compile-time so the sweep's
 // per-thread granule count (2*CG) is a static register array.
 template <int CG>
@@ -48,14 +68,14 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                    float* __restrict__ c_seq,                   // [T,B,H]
                    unsigned long long* __restrict__ hx,         // [2][RT][16][H/2] zeroed
                    unsigned int* __restrict__ err,
-                   int T, int B, int H, int stash) {
+                   int T, int B, int H, int stash, int xcd_swz) {
     extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x h tile [16][H] bf16
 
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
     const int lane = tid & 63;
-    const int rt = blockIdx.x;            // row tile
-    const int cg = blockIdx.y;            // column group
+    int rt, cg;                           // row tile / column group
+    px_decode_block<CG>(xcd_swz, rt, cg);
     const int brow0 = rt * 16;
     const int HQg = H / CG;               // h-cols this block owns
     const int col_base = cg * HQg + wid * 16;  // one 16-col fragment per wave
@@ -230,6 +250,14 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
 // budget returns -2 so the caller serializes it behind the in-flight px
 // grids instead of risking a co-residency livelock. Success returns the
 // grid's CU-equivalents (positive) for the caller to commit.
+static inline int px_xcd_swz() {
+    static const int v = []() {
+        const char* e = getenv("R2D2_PX_NO_XCD_SWZ");
+        return (e && e[0] && e[0] != '0') ? 0 : 1;
+    }();
+    return v;
+}
+
 static inline int px_cu_eq(int blocks, int waves_pb, long lds_bytes) {
     int slots = (int)(160L * 1024 / (lds_bytes > 0 ? lds_bytes : 1));
     int wslots = 32 / (waves_pb > 0 ? waves_pb : 1);
@@ -268,21 +296,24 @@ extern "C" int launch_lstm_fwd_px_bf16(
                 (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
                 (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
                 (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
-                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash,
+                px_xcd_swz());
             break;
         case 4:
             lstm_fwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(
                 (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
                 (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
                 (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
-                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash,
+                px_xcd_swz());
             break;
         case 8:
             lstm_fwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(
                 (const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
                 (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
                 (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
-                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash);
+                (__hip_bfloat16*)gates_act, c_seq, hx, err, T, B, H, stash,
+                px_xcd_swz());
             break;
         default:
             return -1;
@@ -314,6 +345,7 @@ struct BwdPxArgs {
     unsigned char* slab;             // [2][RT][16*4H*2] bytes
     unsigned long long* flags;       // [2][RT][CG] zeroed
     int T;
+    int xcd_swz;
 };
 
 template <int CG>
@@ -336,8 +368,8 @@ DEV_INLINE void lstm_bwd_px_body(const BwdPxArgs& a, unsigned int* err,
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int rt = blockIdx.x;
-    const int cg = blockIdx.y;
+    int rt, cg;
+    px_decode_block<CG>(a.xcd_swz, rt, cg);
     const int brow0 = rt * 16;
     const int HQg = H / CG;
     const int col_base = cg * HQg + wid * 16;
@@ -590,7 +622,8 @@ extern "C" int launch_lstm_bwd_px_bf16(
                 (const __hip_bfloat16*)dcT, (const __hip_bfloat16*)gates_act,
                 c_seq, (const __hip_bfloat16*)c0,
                 (const __hip_bfloat16*)whT_packed, (__hip_bfloat16*)dgates,
-                (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0, slab, flags, T};
+                (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0, slab, flags, T,
+                px_xcd_swz()};
     switch (CG) {
         case 2: lstm_bwd_px_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
         case 4: lstm_bwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
@@ -633,12 +666,12 @@ extern "C" int launch_lstm_bwd_px2_bf16(
                  (const __hip_bfloat16*)dcT0, (const __hip_bfloat16*)ga0, cs0,
                  (const __hip_bfloat16*)c00, (const __hip_bfloat16*)whT0,
                  (__hip_bfloat16*)dg0, (__hip_bfloat16*)dh00,
-                 (__hip_bfloat16*)dc00, slab0, flags0, T0};
+                 (__hip_bfloat16*)dc00, slab0, flags0, T0, px_xcd_swz()};
     BwdPxArgs a1{(const __hip_bfloat16*)dh_seq1, (const __hip_bfloat16*)dhT1,
                  (const __hip_bfloat16*)dcT1, (const __hip_bfloat16*)ga1, cs1,
                  (const __hip_bfloat16*)c01, (const __hip_bfloat16*)whT1,
                  (__hip_bfloat16*)dg1, (__hip_bfloat16*)dh01,
-                 (__hip_bfloat16*)dc01, slab1, flags1, T1};
+                 (__hip_bfloat16*)dc01, slab1, flags1, T1, px_xcd_swz()};
     switch (CG) {
         case 2: lstm_bwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
         case 4: lstm_bwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
@@ -669,6 +702,7 @@ struct PxArgs {
     unsigned long long* hx;
     int T;
     int stash;
+    int xcd_swz;
 };
 
 template <int CG>
@@ -681,8 +715,8 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
     const int lane = tid & 63;
-    const int rt = blockIdx.x;
-    const int cg = blockIdx.y;
+    int rt, cg;
+    px_decode_block<CG>(a.xcd_swz, rt, cg);
     const int brow0 = rt * 16;
     const int HQg = H / CG;
     const int col_base = cg * HQg + wid * 16;
@@ -862,11 +896,11 @@ extern "C" int launch_lstm_fwd_px2_bf16(
     PxArgs a0{(const __hip_bfloat16*)gx0, (const __hip_bfloat16*)h00,
               (const __hip_bfloat16*)c00, (const __hip_bfloat16*)wh0,
               (__hip_bfloat16*)hs0, (__hip_bfloat16*)cf0,
-              (__hip_bfloat16*)ga0, cs0, hx0, T0, stash0};
+              (__hip_bfloat16*)ga0, cs0, hx0, T0, stash0, px_xcd_swz()};
     PxArgs a1{(const __hip_bfloat16*)gx1, (const __hip_bfloat16*)h01,
               (const __hip_bfloat16*)c01, (const __hip_bfloat16*)wh1,
               (__hip_bfloat16*)hs1, (__hip_bfloat16*)cf1,
-              (__hip_bfloat16*)ga1, cs1, hx1, T1, stash1};
+              (__hip_bfloat16*)ga1, cs1, hx1, T1, stash1, px_xcd_swz()};
     switch (CG) {
         case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
         case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;

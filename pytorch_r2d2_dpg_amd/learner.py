@@ -285,32 +285,55 @@ class Learner:
 
         # ---- shared tail: all-reduce -> clip -> join -> optimizer steps ----
         cparams = list(self.critic.parameters())
-        self.dist.allreduce_grads(cparams, key="critic")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
-        # the side stream reads critic params: join before the optimizer writes
-        if join_ev is not None:
-            main.wait_event(join_ev)
-        self.opt_critic.step()
+        fuse_buckets = self.dist.world_size > 1
 
-        self._pe("critic", 1)
-        self._pe("actor_opt", 0)
-        # ---- actor optimizer (grads computed on the side stream)
-        for p, g in zip(self._actor_params, actor_grads):
-            if overlap:
-                # g was allocated on the side stream but is consumed by the
-                # all-reduce/optimizer on main: mark the cross-stream use so
-                # the caching allocator does not recycle g for a later
-                # side-stream allocation while main's reads are in flight
-                g.record_stream(main)
-            if p.grad is None:
-                p.grad = g.clone() if not zero_to_none else g
-            else:
-                p.grad.copy_(g)
-        self.dist.allreduce_grads(self._actor_params, key="actor")
-        if cfg.learner.max_grad_norm > 0:
-            torch.nn.utils.clip_grad_norm_(self._actor_params, cfg.learner.max_grad_norm)
-        self.opt_actor.step()
+        def _fill_actor_grads():
+            for p, g in zip(self._actor_params, actor_grads):
+                if overlap:
+                    # g was allocated on the side stream but is consumed by
+                    # the all-reduce/optimizer on main: mark the cross-stream
+                    # use so the caching allocator does not recycle g for a
+                    # later side-stream allocation while main's reads are in
+                    # flight
+                    g.record_stream(main)
+                if p.grad is None:
+                    p.grad = g.clone() if not zero_to_none else g
+                else:
+                    p.grad.copy_(g)
+
+        if fuse_buckets:
+            # DP: ONE fused collective per step. The grad bytes are ~2-20 MB
+            # bf16 — firmly RCCL's latency regime on the xGMI mesh — so one
+            # bucket for critic+actor beats two back-to-back collectives.
+            # Requires the actor grads first -> join the side stream now.
+            if join_ev is not None:
+                main.wait_event(join_ev)
+            _fill_actor_grads()
+            self.dist.allreduce_grads(cparams + self._actor_params, key="all")
+            if cfg.learner.max_grad_norm > 0:
+                torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
+            self.opt_critic.step()
+            self._pe("critic", 1)
+            self._pe("actor_opt", 0)
+            if cfg.learner.max_grad_norm > 0:
+                torch.nn.utils.clip_grad_norm_(self._actor_params,
+                                               cfg.learner.max_grad_norm)
+            self.opt_actor.step()
+        else:
+            if cfg.learner.max_grad_norm > 0:
+                torch.nn.utils.clip_grad_norm_(cparams, cfg.learner.max_grad_norm)
+            # the side stream reads critic params: join before the optimizer
+            # writes
+            if join_ev is not None:
+                main.wait_event(join_ev)
+            self.opt_critic.step()
+            self._pe("critic", 1)
+            self._pe("actor_opt", 0)
+            _fill_actor_grads()
+            if cfg.learner.max_grad_norm > 0:
+                torch.nn.utils.clip_grad_norm_(self._actor_params,
+                                               cfg.learner.max_grad_norm)
+            self.opt_actor.step()
 
         # ---- Polyak soft target update (K4: one fused launch per net on GPU)
         # (branch is Python-side: under graph capture target_update_every == 1)

@@ -250,6 +250,17 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
 // budget returns -2 so the caller serializes it behind the in-flight px
 // grids instead of risking a co-residency livelock. Success returns the
 // grid's CU-equivalents (positive) for the caller to commit.
+// R2D2_PX_CG_MAX caps the column-group fan-out (tuning knob: the best CG
+// shifted when the XCD swizzle made Wh L2-resident — re-measure per shape).
+static inline int px_cg_max() {
+    static const int v = []() {
+        const char* e = getenv("R2D2_PX_CG_MAX");
+        int x = e ? atoi(e) : 8;
+        return (x == 2 || x == 4 || x == 8) ? x : 8;
+    }();
+    return v;
+}
+
 static inline int px_xcd_swz() {
     static const int v = []() {
         const char* e = getenv("R2D2_PX_NO_XCD_SWZ");
@@ -277,6 +288,7 @@ extern "C" int launch_lstm_fwd_px_bf16(
     if (budget > 224) budget = 224;
     int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
+        if (c > px_cg_max()) continue;
         if (H % (16 * c)) continue;          // wave owns 16 cols
         int w = (H / c) / 16;                // waves per block
         if (w < 2 || w > 16) continue;
@@ -601,6 +613,7 @@ extern "C" int launch_lstm_bwd_px_bf16(
     if (budget > 224) budget = 224;
     int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
+        if (c > px_cg_max()) continue;
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
@@ -648,6 +661,7 @@ extern "C" int launch_lstm_bwd_px2_bf16(
     if (budget > 224) budget = 224;
     int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
+        if (c > px_cg_max()) continue;
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;
@@ -880,6 +894,7 @@ extern "C" int launch_lstm_fwd_px2_bf16(
     if (budget > 224) budget = 224;
     int CG = 0, cu_eq = 0;
     for (int c : {8, 4, 2}) {
+        if (c > px_cg_max()) continue;
         if (H % (16 * c)) continue;
         int w = (H / c) / 16;
         if (w < 2 || w > 16) continue;

@@ -601,6 +601,232 @@ lstm_bwd_px2_kernel(BwdPxArgs a0, BwdPxArgs a1, unsigned int* err, int B, int H)
     lstm_bwd_px_body<CG>(blockIdx.z ? a1 : a0, err, B, H, smem);
 }
 
+
+// ---- granule-exchange backward body (default; R2D2_PX_BWD_SLAB=1 selects
+// the slab body above for A/B). dgates are published as tagged 8-byte
+// granules (2 adjacent bf16 columns; tag = T-t; "the data IS the flag" —
+// relaxed agent-scope atomics, exactly the forward's proven pattern), so
+// the per-timestep release chain shrinks from
+//   sc1 stores -> s_waitcnt vmcnt(0) -> __syncthreads -> flag ->
+//   peer flag poll -> slab sweep -> __syncthreads
+// to a direct per-granule poll overlapped with the fire-and-forget global
+// dgates stores: 2 barriers per timestep instead of 5 and no vmcnt drain.
+// Workspace: a.slab reinterpreted as u64 ex[2][RT][CG][16*4*HQg/2]
+// (identical byte size to the slab image; MUST be zero-initialized so
+// garbage can never alias a valid tag). a.flags is unused here.
+// Double-buffer lifetime argument: identical to the forward's (a producer
+// can only reach epoch parity p again after every consumer of that parity
+// finished its previous use — docs/KERNELS.md).
+template <int CG>
+DEV_INLINE void lstm_bwd_px_body_gx(const BwdPxArgs& a, unsigned int* err,
+                                    int B, int H, char* smem) {
+    const __hip_bfloat16* __restrict__ dh_seq = a.dh_seq;
+    const __hip_bfloat16* __restrict__ dhT = a.dhT;
+    const __hip_bfloat16* __restrict__ dcT = a.dcT;
+    const __hip_bfloat16* __restrict__ gates_act = a.gates_act;
+    const float* __restrict__ c_seq = a.c_seq;
+    const __hip_bfloat16* __restrict__ c0 = a.c0;
+    const __hip_bfloat16* __restrict__ whT = a.whT;
+    __hip_bfloat16* __restrict__ dgates = a.dgates;
+    __hip_bfloat16* __restrict__ dh0 = a.dh0;
+    __hip_bfloat16* __restrict__ dc0 = a.dc0;
+    const int T = a.T;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    int rt, cg;
+    px_decode_block<CG>(a.xcd_swz, rt, cg);
+    const int brow0 = rt * 16;
+    const int HQg = H / CG;
+    const int col_base = cg * HQg + wid * 16;
+    const int H4 = 4 * H;
+    const int rowbytes4 = H4 * 2;
+
+    const int a_row = lane & 15;
+    const int a_kgrp = (lane >> 4) * 8;
+    const int c_col = lane & 15;
+    const int c_row0 = (lane >> 4) * 4;
+
+    float dh_reg[4], dc_reg[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        bool ok = grow < B;
+        dh_reg[r] = ok ? bf2f(dhT[(long)grow * H + col]) : 0.0f;
+        dc_reg[r] = ok ? bf2f(dcT[(long)grow * H + col]) : 0.0f;
+    }
+
+    gu64* ex = (gu64*)a.slab;                 // [2][RT][CG][reg_sz]
+    const int hq2 = HQg >> 1;                 // col-pairs per gate segment
+    const long reg_sz = 16L * 4 * hq2;        // granules per (rt, cg) region
+    const long rt_sz = (long)CG * reg_sz;
+    const long epoch_sz = (long)gridDim.x * rt_sz;
+    const int bd = (int)blockDim.x;           // reg_sz / bd == 8 always
+
+    for (int t = T - 1; t >= 0; --t) {
+        const unsigned tag = (unsigned)(T - t);
+        char* buf = smem;                     // single LDS dg tile (see slab body)
+        const __hip_bfloat16* dh_t = dh_seq + (long)t * B * H;
+        const __hip_bfloat16* ga_t = gates_act + (long)t * B * H4;
+        const float* cs_t = c_seq + (long)t * B * H;
+        const float* cs_p = (t > 0) ? c_seq + (long)(t - 1) * B * H : nullptr;
+        __hip_bfloat16* dg_t = dgates + (long)t * B * H4;
+        gu64* ep = ex + (t & 1) * epoch_sz + (long)rt * rt_sz;
+
+        // ---- own dgates: registers -> LDS own region + fire-and-forget
+        // global write + tagged granule publish (no drain, no flag)
+        {
+            const int col = col_base + c_col;
+            const int cp_base = (wid * 16 + c_col) >> 1;
+            gu64* own = ep + (long)cg * reg_sz;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = c_row0 + r;
+                int grow = brow0 + row;
+                float dgi = 0.f, dgf = 0.f, dgg = 0.f, dgo = 0.f;
+                if (grow < B) {
+                    float dht = dh_reg[r] + bf2f(dh_t[(long)grow * H + col]);
+                    float gi = bf2f(ga_t[(long)grow * H4 + 0 * H + col]);
+                    float gf = bf2f(ga_t[(long)grow * H4 + 1 * H + col]);
+                    float gg = bf2f(ga_t[(long)grow * H4 + 2 * H + col]);
+                    float go = bf2f(ga_t[(long)grow * H4 + 3 * H + col]);
+                    float ct = cs_t[(long)grow * H + col];
+                    float cp2 = (t > 0) ? cs_p[(long)grow * H + col]
+                                        : bf2f(c0[(long)grow * H + col]);
+                    float tc = px_tanh(ct);
+                    float do_ = dht * tc;
+                    float dc = dht * go * (1.0f - tc * tc) + dc_reg[r];
+                    dgi = dc * gg * gi * (1.0f - gi);
+                    dgf = dc * cp2 * gf * (1.0f - gf);
+                    dgg = dc * gi * (1.0f - gg * gg);
+                    dgo = do_ * go * (1.0f - go);
+                    dc_reg[r] = dc * gf;
+                } else {
+                    dc_reg[r] = 0.0f;
+                }
+                const float vals[4] = {dgi, dgf, dgg, dgo};
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+                    unsigned short hb = f32_to_bf16_bits(vals[g]);
+                    *reinterpret_cast<__hip_bfloat16*>(
+                        &buf[px_swz(row * rowbytes4 + (g * H + col) * 2, row)]) =
+                        f2bf(vals[g]);
+                    if (grow < B)
+                        dg_t[(long)grow * H4 + g * H + col] = f2bf(vals[g]);
+                    unsigned other = __shfl_xor((unsigned)hb, 1);
+                    if ((lane & 1) == 0) {
+                        unsigned lo = (unsigned)hb | (other << 16);
+                        __hip_atomic_store(
+                            &own[(long)row * (4 * hq2) + g * hq2 + cp_base],
+                            ((unsigned long long)tag << 32) | lo, RLX_AGENT);
+                    }
+                }
+            }
+        }
+
+        // ---- sweep peers' granules into the LDS tile, one peer region per
+        // batch (8 granules/thread/peer; batched loads, re-poll stragglers)
+        for (int pofs = 1; pofs < CG; ++pofs) {
+            int p = cg + pofs;
+            if (p >= CG) p -= CG;             // stagger start peer per block
+            gu64* src = ep + (long)p * reg_sz;
+            unsigned long long v[8];
+            unsigned pending = 0;
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                v[i] = __hip_atomic_load(&src[tid + i * bd], RLX_AGENT);
+                pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
+            }
+            unsigned spins = 0;
+            while (pending) {
+                if (++spins > 20000000u) {  // convergent bail (see fwd)
+                    __hip_atomic_store((gu32*)err, 0x20000u + (unsigned)t,
+                                       RLX_AGENT);
+                    break;
+                }
+                __builtin_amdgcn_s_sleep(1);
+#pragma unroll
+                for (int i = 0; i < 8; ++i)
+                    if (pending & (1u << i)) {
+                        unsigned long long nv =
+                            __hip_atomic_load(&src[tid + i * bd], RLX_AGENT);
+                        if ((unsigned)(nv >> 32) == tag) {
+                            v[i] = nv;
+                            pending &= ~(1u << i);
+                        }
+                    }
+            }
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                int e = tid + i * bd;
+                int row = e / (4 * hq2);
+                int rem = e - row * (4 * hq2);
+                int g = rem / hq2;
+                int cpx = rem - g * hq2;
+                unsigned off = px_swz(
+                    (unsigned)(row * rowbytes4 + (g * H + p * HQg + cpx * 2) * 2),
+                    row);
+                *reinterpret_cast<unsigned*>(&buf[off]) = (unsigned)v[i];
+            }
+        }
+        __syncthreads();
+
+        // ---- dh recurrence (4-accumulator ILP split, as the slab body)
+        f32x4 acc4[4] = {};
+        const int nb = (col_base >> 4);
+        for (int kk = 0; kk < H4; kk += 128) {
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                int kx = kk + u * 32;
+                int k = kx + a_kgrp;
+                short8 a_frag = *reinterpret_cast<const short8*>(
+                    &buf[px_swz(a_row * rowbytes4 + k * 2, a_row)]);
+                long off = (((long)nb * (H4 >> 5) + (kx >> 5)) * 64 + lane) * 8;
+                short8 b_frag = *reinterpret_cast<const short8*>(&whT[off]);
+                acc4[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, b_frag, acc4[u], 0, 0, 0);
+            }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            dh_reg[r] = acc4[0][r] + acc4[1][r] + acc4[2][r] + acc4[3][r];
+        __syncthreads();  // MFMA reads done before next step's LDS writes
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        if (grow < B) {
+            dh0[(long)grow * H + col] = f2bf(dh_reg[r]);
+            dc0[(long)grow * H + col] = f2bf(dc_reg[r]);
+        }
+    }
+}
+
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_bwd_px_kernel_gx(BwdPxArgs a, unsigned int* err, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    lstm_bwd_px_body_gx<CG>(a, err, B, H, smem);
+}
+
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_bwd_px2_kernel_gx(BwdPxArgs a0, BwdPxArgs a1, unsigned int* err, int B, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    lstm_bwd_px_body_gx<CG>(blockIdx.z ? a1 : a0, err, B, H, smem);
+}
+
+static inline int px_bwd_slab() {
+    static const int v = []() {
+        const char* e = getenv("R2D2_PX_BWD_SLAB");
+        return (e && e[0] && e[0] != '0') ? 1 : 0;
+    }();
+    return v;
+}
+
 extern "C" int launch_lstm_bwd_px_bf16(
     const void* dh_seq, const void* dhT, const void* dcT, const void* gates_act,
     const float* c_seq, const void* c0, const void* whT_packed,
@@ -637,11 +863,20 @@ extern "C" int launch_lstm_bwd_px_bf16(
                 (const __hip_bfloat16*)whT_packed, (__hip_bfloat16*)dgates,
                 (__hip_bfloat16*)dh0, (__hip_bfloat16*)dc0, slab, flags, T,
                 px_xcd_swz()};
-    switch (CG) {
-        case 2: lstm_bwd_px_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
-        case 4: lstm_bwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
-        case 8: lstm_bwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
-        default: return -1;
+    if (px_bwd_slab()) {
+        switch (CG) {
+            case 2: lstm_bwd_px_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            case 4: lstm_bwd_px_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            case 8: lstm_bwd_px_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            default: return -1;
+        }
+    } else {
+        switch (CG) {
+            case 2: lstm_bwd_px_kernel_gx<2><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            case 4: lstm_bwd_px_kernel_gx<4><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            case 8: lstm_bwd_px_kernel_gx<8><<<grid, dim3(waves * 64), lds, stream>>>(a, err, B, H); break;
+            default: return -1;
+        }
     }
     LSTM_CHECK_LAUNCH();
     return cu_eq;
@@ -686,11 +921,20 @@ extern "C" int launch_lstm_bwd_px2_bf16(
                  (const __hip_bfloat16*)c01, (const __hip_bfloat16*)whT1,
                  (__hip_bfloat16*)dg1, (__hip_bfloat16*)dh01,
                  (__hip_bfloat16*)dc01, slab1, flags1, T1, px_xcd_swz()};
-    switch (CG) {
-        case 2: lstm_bwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        case 4: lstm_bwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        case 8: lstm_bwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        default: return -1;
+    if (px_bwd_slab()) {
+        switch (CG) {
+            case 2: lstm_bwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 4: lstm_bwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 8: lstm_bwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            default: return -1;
+        }
+    } else {
+        switch (CG) {
+            case 2: lstm_bwd_px2_kernel_gx<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 4: lstm_bwd_px2_kernel_gx<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 8: lstm_bwd_px2_kernel_gx<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            default: return -1;
+        }
     }
     LSTM_CHECK_LAUNCH();
     return cu_eq;

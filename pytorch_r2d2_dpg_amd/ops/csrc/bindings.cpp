@@ -437,8 +437,9 @@ lstm_seq_bwd(torch::Tensor dh_seq, torch::Tensor dhT, torch::Tensor dcT,
     if (dt == torch::kBFloat16) {
         const int RT = (int)((B + 15) / 16);
         // ZEROS, not empty: the granule-exchange backward reinterprets this
-        // as tagged u64 granules — garbage must never alias a valid tag
-        auto slab = torch::zeros({2LL, RT, 16 * 4 * H * 2},
+        // as tagged u64 granules (8 B per 2 bf16 payload = 2x the slab
+        // image) — and garbage must never alias a valid tag
+        auto slab = torch::zeros({2LL, RT, 16 * 4 * H * 4},
                                  dh_seq.options().dtype(torch::kUInt8));
         auto flags = torch::zeros({2LL, RT, 8}, dh_seq.options().dtype(torch::kInt64));
         auto& err = px_err_word(dh_seq.device());
@@ -510,9 +511,9 @@ lstm_seq_bwd_pair(torch::Tensor dh0s, torch::Tensor dhT0, torch::Tensor dcT0,
     auto dh01 = torch::empty({B, H}, opts);
     auto dc00 = torch::empty({B, H}, opts);
     auto dc01 = torch::empty({B, H}, opts);
-    // zeros: see lstm_seq_bwd (tagged-granule workspace)
-    auto slab0 = torch::zeros({2LL, RT, 16 * 4 * H * 2}, opts.dtype(torch::kUInt8));
-    auto slab1 = torch::zeros({2LL, RT, 16 * 4 * H * 2}, opts.dtype(torch::kUInt8));
+    // zeros: see lstm_seq_bwd (tagged-granule workspace, 2x the slab image)
+    auto slab0 = torch::zeros({2LL, RT, 16 * 4 * H * 4}, opts.dtype(torch::kUInt8));
+    auto slab1 = torch::zeros({2LL, RT, 16 * 4 * H * 4}, opts.dtype(torch::kUInt8));
     auto flags0 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
     auto flags1 = torch::zeros({2LL, RT, 8}, opts.dtype(torch::kInt64));
     auto& err = px_err_word(dh0s.device());

@@ -602,8 +602,10 @@ lstm_bwd_px2_kernel(BwdPxArgs a0, BwdPxArgs a1, unsigned int* err, int B, int H)
 }
 
 
-// ---- granule-exchange backward body (default; R2D2_PX_BWD_SLAB=1 selects
-// the slab body above for A/B). dgates are published as tagged 8-byte
+// ---- granule-exchange backward body (opt-in via R2D2_PX_BWD_GX=1;
+// measured SLOWER than the slab body — see px_bwd_slab() — and kept as the
+// documented negative result + a second implementation of the exchange
+// contract). dgates are published as tagged 8-byte
 // granules (2 adjacent bf16 columns; tag = T-t; "the data IS the flag" —
 // relaxed agent-scope atomics, exactly the forward's proven pattern), so
 // the per-timestep release chain shrinks from
@@ -819,10 +821,15 @@ lstm_bwd_px2_kernel_gx(BwdPxArgs a0, BwdPxArgs a1, unsigned int* err, int B, int
     lstm_bwd_px_body_gx<CG>(blockIdx.z ? a1 : a0, err, B, H, smem);
 }
 
+// MEASURED on MI355X (flagship, probe triplets): slab 4.95 ms/step vs
+// granules 5.18 — the slab's vectorized 16B bulk stores/loads beat the
+// per-granule 8B atomic traffic (2x bytes for tags, 7 sequential per-peer
+// poll rounds) despite paying the vmcnt drain + flag chain. The slab body
+// stays the DEFAULT; R2D2_PX_BWD_GX=1 selects the granule body.
 static inline int px_bwd_slab() {
     static const int v = []() {
-        const char* e = getenv("R2D2_PX_BWD_SLAB");
-        return (e && e[0] && e[0] != '0') ? 1 : 0;
+        const char* e = getenv("R2D2_PX_BWD_GX");
+        return (e && e[0] && e[0] != '0') ? 0 : 1;
     }();
     return v;
 }

@@ -132,7 +132,8 @@ class Learner:
         # serializes behind the in-flight grid, so concurrent px can no
         # longer livelock. R2D2_NO_OVERLAP=1 forces off.
         overlap_ok = (self.device.type == "cuda"
-                      and cfg.learner.batch_size <= 256
+                      and (cfg.learner.batch_size <= 256
+                           or bool(os.environ.get("R2D2_FORCE_OVERLAP")))
                       and not os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()

@@ -121,19 +121,15 @@ class Learner:
         # stream-overlap machinery (GPU only): the actor chain runs on a side
         # stream concurrently with the target/critic chain
         self._actor_params = list(self.actor.parameters())
-        # Stream overlap pays at B <= 256 (glue hides behind the LSTM
-        # chains, +20% measured); at larger batches two concurrent
-        # column-split grids contend for CU residency and crawl, so run the
-        # chains sequentially there (the px kernels fill the chip anyway).
-        # The r1 train_len <= 48 gate (seq-120 spin-outs when two bwd-px
-        # grids were an exact 256-CU fit) is GONE: the px launchers now do
-        # host-side CU-budget accounting (bindings.cpp px_budget_*) — a
-        # launch that would over-subscribe the chip picks a smaller CG or
-        # serializes behind the in-flight grid, so concurrent px can no
-        # longer livelock. R2D2_NO_OVERLAP=1 forces off.
+        # Stream overlap + pairing is the default at EVERY batch size now.
+        # Round 1 gated it to B <= 256 and train_len <= 48 after observing
+        # CU-residency crawls/livelocks between concurrent column-split
+        # grids; both failure modes are gone under the round-2 launcher
+        # (CU-budget accounting with drain-on-over-subscription,
+        # bindings.cpp px_budget_*) — measured: seq-120 B=256 +42% and
+        # B=512/seq-120 +31% from lifting the gates, zero px spin
+        # timeouts. R2D2_NO_OVERLAP=1 forces the sequential chains.
         overlap_ok = (self.device.type == "cuda"
-                      and (cfg.learner.batch_size <= 256
-                           or bool(os.environ.get("R2D2_FORCE_OVERLAP")))
                       and not os.environ.get("R2D2_NO_OVERLAP"))
         if overlap_ok:
             self._side_stream = torch.cuda.Stream()

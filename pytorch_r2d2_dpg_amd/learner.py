@@ -19,13 +19,14 @@ contracts for the multi-stream execution are documented in docs/STREAMS.md.
 
 Execution (GPU): independent unrolls share paired (grid.z=2) column-split
 launches — the two burn-ins, the target-actor unroll with the online
-critic forward, and (bf16 + overlap) the two critic-LSTM backwards; the
-actor chain (through the PRE-update critic) runs on a side HIP stream
-overlapped with the target/critic chain at B<=256 and train_len<=48; the
-two gradient all-reduces are issued on the main stream in a fixed order.
-``capture_graph()`` can record the whole update into a hipGraph, but eager
-+ stream overlap measures faster at these sizes (steady-state device
-occupancy is already ~92%), so it is not the default.
+critic forward, and (bf16) the two critic-LSTM backwards; the actor chain
+(through the PRE-update critic) runs on a side HIP stream overlapped with
+the target/critic chain at every batch size (the round-1 B/seq gates are
+gone — the CU-budget px launcher removed the residency livelocks); at
+world>1 the critic+actor gradients go out as ONE fused RCCL bucket on the
+main stream. ``capture_graph()`` can record the whole update into a
+hipGraph, but eager + stream overlap measures faster at these sizes
+(steady-state device occupancy is already ~92%), so it is not the default.
 
 The step is ONE code path (`_step_core`): shared unpack → shared burn-in →
 a pluggable middle (`_mid_pair_bwd` = manual paired-K2 backward, default on

@@ -107,6 +107,7 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     gu64* hx_g = (gu64*)hx;
     const long rt_off = (long)rt * 16 * (H / 2);
     const long slot_sz = (long)gridDim.x * 16 * (H / 2);
+    unsigned spin_limit = 20000000u;  // degrades after the first timeout
 
     for (int t = 0; t < T; ++t) {
         const __hip_bfloat16* gx_t = gates_x + (long)t * B * H4;
@@ -142,12 +143,16 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 20000000u) {
+                if (++spins > spin_limit) {
                     // bounded spin (~0.5 s at s_sleep granularity): flag the error and PROCEED with stale
                     // data — control flow stays convergent so no thread
                     // ever skips a barrier (a diverged exit deadlocks the
-                    // block and hangs the device)
+                    // block and hangs the device). After the FIRST timeout
+                    // the bound degrades to ~1 ms so a genuine co-residency
+                    // violation finishes the kernel fast (T bails) instead
+                    // of spinning 0.5 s per timestep.
                     __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
+                    spin_limit = 40000u;
                     break;
                 }
                 __builtin_amdgcn_s_sleep(1);
@@ -409,6 +414,7 @@ DEV_INLINE void lstm_bwd_px_body(const BwdPxArgs& a, unsigned int* err,
     const long slab_slot = (long)gridDim.x * slab_tile;
     gu64b* flg = (gu64b*)flags;
     const int own_byte_lo = (cg * HQg) * 2;   // within one gate's row segment
+    unsigned spin_limit = 20000000u;          // degrades after first timeout
 
     for (int t = T - 1; t >= 0; --t) {
         const unsigned tag = (unsigned)(T - t);
@@ -516,13 +522,14 @@ DEV_INLINE void lstm_bwd_px_body(const BwdPxArgs& a, unsigned int* err,
                 unsigned long long v = __hip_atomic_load(f, RLX_AGENT);
                 unsigned spins = 0;
                 while ((unsigned)(v >> 32) != tag) {
-                    if (++spins > 20000000u) {  // convergent bail (see fwd)
+                    if (++spins > spin_limit) {  // convergent bail (see fwd)
                         __hip_atomic_store((gu32*)err, 0x10000u + (unsigned)t, RLX_AGENT);
                         break;
                     }
                     __builtin_amdgcn_s_sleep(1);
                     v = __hip_atomic_load(f, RLX_AGENT);
                 }
+                if (spins > spin_limit) spin_limit = 40000u;
             }
             __syncthreads();
             unsigned char* slab_rt = slab + (t & 1) * slab_slot + (long)rt * slab_tile;
@@ -666,6 +673,7 @@ DEV_INLINE void lstm_bwd_px_body_gx(const BwdPxArgs& a, unsigned int* err,
     const long rt_sz = (long)CG * reg_sz;
     const long epoch_sz = (long)gridDim.x * rt_sz;
     const int bd = (int)blockDim.x;           // reg_sz / bd == 8 always
+    unsigned spin_limit = 20000000u;          // degrades after first timeout
 
     for (int t = T - 1; t >= 0; --t) {
         const unsigned tag = (unsigned)(T - t);
@@ -743,9 +751,10 @@ DEV_INLINE void lstm_bwd_px_body_gx(const BwdPxArgs& a, unsigned int* err,
             }
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 20000000u) {  // convergent bail (see fwd)
+                if (++spins > spin_limit) {  // convergent bail (see fwd)
                     __hip_atomic_store((gu32*)err, 0x20000u + (unsigned)t,
                                        RLX_AGENT);
+                    spin_limit = 40000u;
                     break;
                 }
                 __builtin_amdgcn_s_sleep(1);
@@ -1012,6 +1021,7 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
     gu64* hx_g = (gu64*)a.hx;
     const long rt_off = (long)rt * 16 * (H / 2);
     const long slot_sz = (long)gridDim.x * 16 * (H / 2);
+    unsigned spin_limit = 20000000u;  // degrades after the first timeout
 
     for (int t = 0; t < T; ++t) {
         const __hip_bfloat16* gx_t = a.gates_x + (long)t * B * H4;
@@ -1041,8 +1051,9 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
                 pending |= (unsigned)((unsigned)(v[i] >> 32) != tag) << i;
             unsigned spins = 0;
             while (pending) {
-                if (++spins > 20000000u) {  // convergent bail (see single fwd)
+                if (++spins > spin_limit) {  // convergent bail (see single fwd)
                     __hip_atomic_store((gu32*)err, 1u + (unsigned)t, RLX_AGENT);
+                    spin_limit = 40000u;
                     break;
                 }
                 __builtin_amdgcn_s_sleep(1);

@@ -95,6 +95,11 @@ class ActorConfig:
     sigma_alpha: float = 7.0
     eval_actors: int = 0            # noise-free actors for evaluation logging
     max_env_steps: int = 0          # 0 = unbounded
+    torch_threads: int = 1          # intra-op threads per actor process
+                                    # (vectorized actors batch K envs through
+                                    # one CPU forward; >1 thread speeds the
+                                    # policy GEMMs when cores outnumber
+                                    # processes)
 
 
 @dataclass

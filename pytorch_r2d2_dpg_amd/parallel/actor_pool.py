@@ -59,8 +59,8 @@ def _actor_main(actor_id: int, slot: int, cfg_dict: dict, seq_queue, weight_buf,
                 eval_mode: bool = False, ring=None):
     # actor_id is GLOBAL (seeds, sigma ladder); slot is the LOCAL pool index
     # (heartbeat array) — on learner rank r>0 they differ
-    torch.set_num_threads(1)
     cfg = Config.from_dict(cfg_dict)
+    torch.set_num_threads(max(cfg.actor.torch_threads, 1))
     E = max(cfg.actor.envs_per_actor, 1)
     if E > 1 and not eval_mode:
         worker = VectorActorWorker(cfg, actor_id=actor_id, num_envs=E)

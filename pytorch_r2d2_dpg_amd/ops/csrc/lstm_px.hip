@@ -246,6 +246,16 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
     }
 }
 
+// Defined after PxArgs (file order): dispatches the pipelined per-lane
+// forward for the solo entry point. Returns 0 on success, -1 unsupported.
+static int launch_fwd_pl_solo(const void* gates_x, const void* h0,
+                              const void* c0, const void* wh_packed,
+                              void* h_seq, void* c_final, void* gates_act,
+                              float* c_seq, unsigned long long* hx,
+                              unsigned int* err, int T, int B, int H,
+                              int stash, int CG, int waves,
+                              hipStream_t stream);
+
 // CG SELECTION is budget-independent (the r1 rule, RT*CG(*2) <= 224 —
 // measured fastest: budget-driven CG reduction cost the flagship 17%).
 // The budget is a SAFETY NET only: capacity is counted in CU-EQUIVALENTS
@@ -262,6 +272,14 @@ static inline int px_cg_max() {
         const char* e = getenv("R2D2_PX_CG_MAX");
         int x = e ? atoi(e) : 8;
         return (x == 2 || x == 4 || x == 8) ? x : 8;
+    }();
+    return v;
+}
+
+static inline int px_fwd_pl() {
+    static const int v = []() {
+        const char* e = getenv("R2D2_PX_FWD_PL");
+        return (e && e[0] && e[0] != '0') ? 1 : 0;
     }();
     return v;
 }
@@ -305,6 +323,13 @@ extern "C" int launch_lstm_fwd_px_bf16(
     if (CG == 0) return -1;
     if (cu_eq > budget) return -2;           // caller: drain in-flight, retry
     int waves = (H / CG) / 16;
+    if (px_fwd_pl()) {
+        if (launch_fwd_pl_solo(gates_x, h0, c0, wh_packed, h_seq, c_final,
+                               gates_act, c_seq, hx, err, T, B, H, stash,
+                               CG, waves, stream) == 0)
+            return cu_eq;
+        return -1;
+    }
     dim3 grid(RT, CG);
     size_t lds = (size_t)2 * 16 * H * 2;
     switch (CG) {
@@ -1143,6 +1168,200 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
     }
 }
 
+
+// ---- pipelined forward body (opt-in via R2D2_PX_FWD_PL=1): the h exchange
+// feeds MFMA A-FRAGMENTS directly from per-lane granule polls — no LDS
+// staging, no barriers in the T-loop at all. Key observation: lane l of a
+// 16x16x32 MFMA needs exactly rows l&15 x h-cols (l>>4)*8..+8 = 4
+// CONSECUTIVE granules of the exchange slot, and the granule payload
+// (2 adjacent bf16 cols) is already in a_frag order — 4 tag-checked u64
+// loads ARE the fragment. The k-loop polls chunk i while chunk i+1's loads
+// are in flight (2-deep pipeline), so MFMA starts on the first-arrived
+// chunk instead of after the block-wide sweep of the SLOWEST granule, and
+// the two per-step barriers + 16 LDS round-trips per thread disappear.
+// Per-lane divergent spins are safe precisely because there are no
+// barriers; the bail degrades like the other bodies. Costs: waves re-load
+// granules redundantly from L2 (4x traffic on a 16 KB/step tile — cheap)
+// and t=0 reads h0 straight from global.
+template <int CG>
+DEV_INLINE void lstm_fwd_px_body_pl(const PxArgs& a, unsigned int* err,
+                                    int B, int H) {
+    const int T = a.T;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    int rt, cg;
+    px_decode_block<CG>(a.xcd_swz, rt, cg);
+    const int brow0 = rt * 16;
+    const int HQg = H / CG;
+    const int col_base = cg * HQg + wid * 16;
+    const int H4 = 4 * H;
+
+    const int a_row = lane & 15;
+    const int a_kgrp = (lane >> 4) * 8;
+    const int c_col = lane & 15;
+    const int c_row0 = (lane >> 4) * 4;
+
+    float c_reg[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        c_reg[r] = (grow < B) ? bf2f(a.c0[(long)grow * H + col]) : 0.0f;
+    }
+
+    gu64* hx_g = (gu64*)a.hx;
+    const long rt_off = (long)rt * 16 * (H / 2);
+    const long slot_sz = (long)gridDim.x * 16 * (H / 2);
+    unsigned spin_limit = 20000000u;
+    const int a_base_row = a_row * (H >> 1);
+
+    for (int t = 0; t < T; ++t) {
+        const __hip_bfloat16* gx_t = a.gates_x + (long)t * B * H4;
+        f32x4 acc[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int grow = brow0 + c_row0 + r;
+                int j = g * H + col_base + c_col;
+                acc[g][r] = (grow < B) ? bf2f(gx_t[(long)grow * H4 + j]) : 0.0f;
+            }
+
+        const unsigned tag = (unsigned)t;  // consuming h(t-1)
+        gu64* slot = hx_g + ((t - 1) & 1) * slot_sz + rt_off;
+        const int hgrow = brow0 + a_row;
+
+        unsigned long long vc[4], vn[4];
+        if (t > 0) {
+            int base = a_base_row + (a_kgrp >> 1);
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                vc[i] = __hip_atomic_load(&slot[base + i], RLX_AGENT);
+        }
+        for (int kk = 0; kk < H; kk += 32) {
+            // issue next chunk loads while finishing this one
+            if (t > 0 && kk + 32 < H) {
+                int nbase = a_base_row + ((kk + 32 + a_kgrp) >> 1);
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    vn[i] = __hip_atomic_load(&slot[nbase + i], RLX_AGENT);
+            }
+            short8 a_frag;
+            if (t == 0) {
+                if (hgrow < B)
+                    a_frag = *reinterpret_cast<const short8*>(
+                        &a.h0[(long)hgrow * H + kk + a_kgrp]);
+                else
+                    a_frag = short8{0, 0, 0, 0, 0, 0, 0, 0};
+            } else {
+                int base = a_base_row + ((kk + a_kgrp) >> 1);
+                unsigned pending = 0;
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    pending |= (unsigned)((unsigned)(vc[i] >> 32) != tag) << i;
+                unsigned spins = 0;
+                while (pending) {
+                    if (++spins > spin_limit) {  // per-lane bail, no barriers
+                        __hip_atomic_store((gu32*)err, 1u + (unsigned)t,
+                                           RLX_AGENT);
+                        spin_limit = 40000u;
+                        break;
+                    }
+                    __builtin_amdgcn_s_sleep(1);
+#pragma unroll
+                    for (int i = 0; i < 4; ++i)
+                        if (pending & (1u << i)) {
+                            unsigned long long nv =
+                                __hip_atomic_load(&slot[base + i], RLX_AGENT);
+                            if ((unsigned)(nv >> 32) == tag) {
+                                vc[i] = nv;
+                                pending &= ~(1u << i);
+                            }
+                        }
+                }
+                unsigned u[4];
+#pragma unroll
+                for (int i = 0; i < 4; ++i) u[i] = (unsigned)vc[i];
+                a_frag = *reinterpret_cast<const short8*>(u);
+            }
+            short8 breg[4];
+            const int kb = kk >> 5;
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                int jb = g * (H >> 4) + (col_base >> 4);
+                long off = (((long)jb * (H >> 5) + kb) * 64 + lane) * 8;
+                breg[g] = *reinterpret_cast<const short8*>(&a.wh[off]);
+            }
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+                acc[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag, breg[g], acc[g], 0, 0, 0);
+#pragma unroll
+            for (int i = 0; i < 4; ++i) vc[i] = vn[i];
+        }
+
+        // ---- epilogue: activations, cell update, granule publish, stores
+        {
+            __hip_bfloat16* hs_t = a.h_seq + (long)t * B * H;
+            __hip_bfloat16* ga_t = a.stash ? a.gates_act + (long)t * B * H4 : nullptr;
+            float* cs_t = a.stash ? a.c_seq + (long)t * B * H : nullptr;
+            gu64* oslot = hx_g + (t & 1) * slot_sz + rt_off;
+            const unsigned otag = (unsigned)(t + 1);
+            const int col = col_base + c_col;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = c_row0 + r;
+                int grow = brow0 + row;
+                float gi = px_sigm(acc[0][r]);
+                float gf = px_sigm(acc[1][r]);
+                float gg = px_tanh(acc[2][r]);
+                float go = px_sigm(acc[3][r]);
+                float c = gf * c_reg[r] + gi * gg;
+                float hn = go * px_tanh(c);
+                c_reg[r] = c;
+                unsigned short hb = f32_to_bf16_bits(hn);
+                unsigned other = __shfl_xor((unsigned)hb, 1);
+                if ((lane & 1) == 0) {
+                    unsigned lo = (unsigned)hb | (other << 16);
+                    __hip_atomic_store(&oslot[row * (H / 2) + col / 2],
+                                       ((unsigned long long)otag << 32) | lo,
+                                       RLX_AGENT);
+                }
+                if (grow < B) {
+                    hs_t[(long)grow * H + col] = f2bf(hn);
+                    if (a.stash) {
+                        cs_t[(long)grow * H + col] = c;
+                        ga_t[(long)grow * H4 + 0 * H + col] = f2bf(gi);
+                        ga_t[(long)grow * H4 + 1 * H + col] = f2bf(gf);
+                        ga_t[(long)grow * H4 + 2 * H + col] = f2bf(gg);
+                        ga_t[(long)grow * H4 + 3 * H + col] = f2bf(go);
+                    }
+                }
+            }
+        }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int grow = brow0 + c_row0 + r;
+        int col = col_base + c_col;
+        if (grow < B) a.c_final[(long)grow * H + col] = f2bf(c_reg[r]);
+    }
+}
+
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_fwd_px_pl_kernel(PxArgs a, unsigned int* err, int B, int H) {
+    lstm_fwd_px_body_pl<CG>(a, err, B, H);
+}
+
+template <int CG>
+__global__ void __launch_bounds__(CG >= 8 ? 512 : 1024)
+lstm_fwd_px2_pl_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
+    lstm_fwd_px_body_pl<CG>(blockIdx.z ? a1 : a0, err, B, H);
+}
+
+
 extern "C" int launch_lstm_fwd_px2_bf16(
     const void* gx0, const void* h00, const void* c00, const void* wh0,
     void* hs0, void* cf0, void* ga0, float* cs0, int stash0, int T0,
@@ -1178,12 +1397,45 @@ extern "C" int launch_lstm_fwd_px2_bf16(
               (const __hip_bfloat16*)c01, (const __hip_bfloat16*)wh1,
               (__hip_bfloat16*)hs1, (__hip_bfloat16*)cf1,
               (__hip_bfloat16*)ga1, cs1, hx1, T1, stash1, px_xcd_swz()};
-    switch (CG) {
-        case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
-        default: return -1;
+    if (px_fwd_pl()) {
+        switch (CG) {
+            case 2: lstm_fwd_px2_pl_kernel<2><<<grid, dim3(waves * 64), 0, stream>>>(a0, a1, err, B, H); break;
+            case 4: lstm_fwd_px2_pl_kernel<4><<<grid, dim3(waves * 64), 0, stream>>>(a0, a1, err, B, H); break;
+            case 8: lstm_fwd_px2_pl_kernel<8><<<grid, dim3(waves * 64), 0, stream>>>(a0, a1, err, B, H); break;
+            default: return -1;
+        }
+    } else {
+        switch (CG) {
+            case 2: lstm_fwd_px2_kernel<2><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 4: lstm_fwd_px2_kernel<4><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            case 8: lstm_fwd_px2_kernel<8><<<grid, dim3(waves * 64), lds, stream>>>(a0, a1, err, B, H); break;
+            default: return -1;
+        }
     }
     LSTM_CHECK_LAUNCH();
     return cu_eq;
+}
+
+// ---- pipelined-forward solo dispatch (see forward declaration above) ------
+static int launch_fwd_pl_solo(const void* gates_x, const void* h0,
+                              const void* c0, const void* wh_packed,
+                              void* h_seq, void* c_final, void* gates_act,
+                              float* c_seq, unsigned long long* hx,
+                              unsigned int* err, int T, int B, int H,
+                              int stash, int CG, int waves,
+                              hipStream_t stream) {
+    int RT = (B + 15) / 16;
+    dim3 grid(RT, CG);
+    PxArgs a{(const __hip_bfloat16*)gates_x, (const __hip_bfloat16*)h0,
+             (const __hip_bfloat16*)c0, (const __hip_bfloat16*)wh_packed,
+             (__hip_bfloat16*)h_seq, (__hip_bfloat16*)c_final,
+             (__hip_bfloat16*)gates_act, c_seq, hx, T, stash, px_xcd_swz()};
+    switch (CG) {
+        case 2: lstm_fwd_px_pl_kernel<2><<<grid, dim3(waves * 64), 0, stream>>>(a, err, B, H); break;
+        case 4: lstm_fwd_px_pl_kernel<4><<<grid, dim3(waves * 64), 0, stream>>>(a, err, B, H); break;
+        case 8: lstm_fwd_px_pl_kernel<8><<<grid, dim3(waves * 64), 0, stream>>>(a, err, B, H); break;
+        default: return -1;
+    }
+    LSTM_CHECK_LAUNCH();
+    return 0;
 }

@@ -564,3 +564,60 @@ def test_rccl_world1_collectives_smoke():
         assert ctx.allreduce_scalar(3.5, op="sum") == 3.5
     finally:
         dist.destroy_process_group()
+
+
+@pytest.mark.gpu
+def test_overlap_scheduling_matches_sequential():
+    """The stream-overlapped step (side-stream actor chain + paired
+    launches) must produce the same post-step parameters as the fully
+    sequential path (R2D2_NO_OVERLAP=1) on an identical batch — scheduling
+    must never change the math."""
+    import os
+    from pytorch_r2d2_dpg_amd.config import Config
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+
+    def mk_cfg():
+        cfg = Config()
+        cfg.env.name = "humanoid-run"
+        cfg.model.hidden_size = 128
+        cfg.model.trunk_sizes = (128, 128)
+        cfg.model.dtype = "bf16"
+        cfg.replay.seq_len = 16
+        cfg.replay.burn_in = 8
+        cfg.learner.batch_size = 24
+        return cfg
+
+    def mk_batch():
+        L = 16 + 5
+        g = torch.Generator().manual_seed(17)
+        return {
+            "obs": torch.randn(24, L, 67, generator=g).cuda(),
+            "action": torch.randn(24, L, 21, generator=g).clamp_(-1, 1).cuda(),
+            "reward": torch.rand(24, L, generator=g).cuda(),
+            "done": torch.zeros(24, L).cuda(),
+            "state": (torch.randn(24, 4, 128, generator=g) * 0.1).cuda(),
+            "weight": torch.ones(24).cuda(),
+            "idx": None,
+        }
+
+    spec = make_env("humanoid-run").spec
+    results = {}
+    for mode in ("overlap", "sequential"):
+        os.environ.pop("R2D2_NO_OVERLAP", None)
+        if mode == "sequential":
+            os.environ["R2D2_NO_OVERLAP"] = "1"
+        torch.manual_seed(5)
+        lrn = Learner(mk_cfg(), spec, device="cuda")
+        assert (mode == "overlap") == (lrn._side_stream is not None)
+        for _ in range(2):
+            lrn.train_step(mk_batch())
+        torch.cuda.synchronize()
+        results[mode] = [p.detach().float().cpu()
+                         for p in list(lrn.actor.parameters())
+                         + list(lrn.critic.parameters())]
+    os.environ.pop("R2D2_NO_OVERLAP", None)
+    assert ext().lstm_px_error() == 0
+    for i, (a, b) in enumerate(zip(results["overlap"], results["sequential"])):
+        assert torch.allclose(a, b, atol=2e-3, rtol=1e-2), \
+            f"param {i} max diff {(a - b).abs().max()}"

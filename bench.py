@@ -113,10 +113,13 @@ def main() -> int:
     p.add_argument("--no-graph", dest="graph", action="store_false")
     args = p.parse_args()
     pixel = args.env == "quadruped-run-pixels"
+    cheetah = args.env == "cheetah-run"
+    # per-env defaults follow the BASELINE.json configs: humanoid (config 3)
+    # B=256/H=512, cheetah (config 2) B=64/H=256, pixels (config 4) B=32/H=256
     if args.batch is None:
-        args.batch = 32 if pixel else 256
+        args.batch = 32 if pixel else (64 if cheetah else 256)
     if args.hidden is None:
-        args.hidden = 256 if pixel else 512
+        args.hidden = 256 if (pixel or cheetah) else 512
     if args.replay_capacity is None:
         args.replay_capacity = 2048 if pixel else 8192
 

@@ -93,6 +93,8 @@ void px_chain_record(hipStream_t st) {
 
 void set_px_serialize(bool on) { g_px_serialize = on; }
 
+bool stream_capturing();  // defined with the pack cache below
+
 // ---- CU-budget accounting for column-split grids ---------------------------
 // Every px kernel requires ALL its blocks co-resident (cross-block spins), so
 // the blocks of all CONCURRENTLY-RUNNING px grids must sum to <= 224 of the
@@ -120,6 +122,9 @@ bool g_px_budget_enabled = []() {
 // the overlap — 6.3 vs 5.2 ms/step.)
 int px_budget_reserve(hipStream_t cur) {
     if (!g_px_budget_enabled) return 224;
+    // during hipGraph capture: event record/query would invalidate the
+    // capture — the graph replays a fixed schedule anyway, so run unbudgeted
+    if (stream_capturing()) return 224;
     // sweep completed entries
     for (size_t i = 0; i < g_px_inflight.size();) {
         if (hipEventQuery(g_px_inflight[i].ev) == hipSuccess) {
@@ -161,6 +166,7 @@ int px_budget_drain(hipStream_t st) {
 void px_budget_commit(int blocks, hipStream_t st) {
     ++g_px_launches;
     if (!g_px_budget_enabled) return;
+    if (stream_capturing()) return;  // see px_budget_reserve
     hipEvent_t ev;
     if (!g_px_event_pool.empty()) {
         ev = g_px_event_pool.back();

@@ -621,3 +621,49 @@ def test_overlap_scheduling_matches_sequential():
     for i, (a, b) in enumerate(zip(results["overlap"], results["sequential"])):
         assert torch.allclose(a, b, atol=2e-3, rtol=1e-2), \
             f"param {i} max diff {(a - b).abs().max()}"
+
+
+@pytest.mark.gpu
+def test_hipgraph_capture_replay():
+    """Learner.capture_graph records the whole update into a hipGraph and
+    replays with fresh batches (regression: the px CU-budget machinery's
+    hipEventRecord during stream capture invalidated the capture)."""
+    import numpy as np
+    from pytorch_r2d2_dpg_amd.config import Config
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+
+    cfg = Config()
+    cfg.env.name = "humanoid-run"
+    cfg.model.hidden_size = 128
+    cfg.model.trunk_sizes = (128, 128)
+    cfg.model.dtype = "bf16"
+    cfg.replay.seq_len = 16
+    cfg.replay.burn_in = 8
+    cfg.learner.batch_size = 16
+
+    def mk_batch(seed):
+        L = 21
+        g = torch.Generator().manual_seed(seed)
+        return {
+            "obs": torch.randn(16, L, 67, generator=g).cuda(),
+            "action": torch.randn(16, L, 21, generator=g).clamp_(-1, 1).cuda(),
+            "reward": torch.rand(16, L, generator=g).cuda(),
+            "done": torch.zeros(16, L).cuda(),
+            "state": (torch.randn(16, 4, 128, generator=g) * 0.1).cuda(),
+            "weight": torch.ones(16).cuda(),
+            "idx": torch.arange(16).cuda(),
+        }
+
+    spec = make_env("humanoid-run").spec
+    torch.manual_seed(3)
+    lrn = Learner(cfg, spec, device="cuda")
+    lrn.capture_graph(mk_batch(0))
+    assert lrn._graph is not None
+    p0 = [p.detach().clone() for p in lrn.actor.parameters()]
+    for s in range(3):
+        st = lrn.train_step(mk_batch(s + 1))
+        assert np.isfinite(st.critic_loss) and np.isfinite(st.actor_loss)
+    torch.cuda.synchronize()
+    assert any(not torch.equal(a, b)
+               for a, b in zip(p0, lrn.actor.parameters())), "replay is a no-op"

@@ -73,6 +73,9 @@ class Learner:
         self.device = torch.device(device)
         self.dist = dist or DistContext.single()
         self.dtype = torch.bfloat16 if cfg.model.dtype == "bf16" else torch.float32
+        if self.device.type == "cuda":
+            from .ops import enable_tunableop
+            enable_tunableop()  # shipped gfx950 GEMM solution table (+4%)
 
         self.actor, self.critic, self.target_actor, self.target_critic = make_models(
             spec, cfg.model, device=self.device, dtype=self.dtype

@@ -50,3 +50,43 @@ def hip_ext():
 def ext_available() -> bool:
     _try_load()
     return _EXT is not None
+
+
+_TUNABLE_DONE = False
+
+
+def enable_tunableop() -> None:
+    """Enable PyTorch TunableOp with the shipped gfx950 GEMM solution table.
+
+    The learner's gates/weight-grad GEMMs are skinny shapes where
+    hipBLASLt's default heuristic picks ~4% off-peak solutions (measured
+    5.00 -> 4.81 ms/step on the flagship); the table was tuned across all
+    five BASELINE configs' shapes on MI355X. Tuning itself stays OFF (the
+    table is read-only; unknown shapes fall back to the heuristic), every
+    DP rank reads the same file, and a version-mismatched install makes
+    TunableOp ignore the table gracefully (validator lines). Opt out with
+    R2D2_NO_TUNABLEOP=1; explicit PYTORCH_TUNABLEOP_* env takes precedence.
+    """
+    global _TUNABLE_DONE
+    if _TUNABLE_DONE:
+        return
+    _TUNABLE_DONE = True
+    import os
+
+    if os.environ.get("R2D2_NO_TUNABLEOP"):
+        return
+    if any(k.startswith("PYTORCH_TUNABLEOP") for k in os.environ):
+        return  # user drives TunableOp themselves (e.g. a tuning run)
+    if not torch.cuda.is_available():
+        return  # set_filename needs a device
+    try:
+        t = torch.cuda.tunable
+    except AttributeError:  # older torch
+        return
+    csv = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                       "tunableop_gfx950.csv")
+    if not os.path.exists(csv):
+        return
+    t.set_filename(csv, insert_device_ordinal=False)
+    t.tuning_enable(False)
+    t.enable(True)

@@ -20,13 +20,6 @@ from torch import nn
 from .lstm import FusedLSTM, State
 
 
-import os as _os
-
-
-def _conv_channels_last() -> bool:
-    return bool(_os.environ.get("R2D2_CONV_CL"))
-
-
 def mlp_trunk(in_dim: int, sizes) -> nn.Sequential:
     layers = []
     d = in_dim
@@ -58,15 +51,7 @@ class ConvTrunk(nn.Module):
             x = x.to(wdtype) / 255.0
         else:
             x = x.to(wdtype)
-        x = x.permute(0, 3, 1, 2)
-        if _conv_channels_last():
-            # NHWC convs (experimental, R2D2_CONV_CL=1): the HWC input is
-            # ALREADY channels_last-contiguous after the permute, and MIOpen
-            # has native NHWC kernels for some shapes
-            x = x.contiguous(memory_format=torch.channels_last)
-        else:
-            x = x.contiguous()
-        return self.net(x)
+        return self.net(x.permute(0, 3, 1, 2).contiguous())
 
 
 class _TrunkEncoder(nn.Module):

@@ -667,3 +667,52 @@ def test_hipgraph_capture_replay():
     torch.cuda.synchronize()
     assert any(not torch.equal(a, b)
                for a, b in zip(p0, lrn.actor.parameters())), "replay is a no-op"
+
+
+@pytest.mark.gpu
+def test_flagship_perf_floor():
+    """Catastrophic-regression guard (SURVEY §4 'performance regression'):
+    the flagship-shaped step must stay far above any eager/fallback path.
+    The floor (1.2M transitions/s) is ~40% under the measured 2.09M so
+    box-to-box variance never trips it, while a silent non-native fallback
+    (~0.3M) or a px livelock fails loudly."""
+    import time
+    from pytorch_r2d2_dpg_amd.config import Config
+    from pytorch_r2d2_dpg_amd.envs import make_env
+    from pytorch_r2d2_dpg_amd.learner import Learner
+
+    cfg = Config()
+    cfg.env.name = "humanoid-run"
+    cfg.model.hidden_size = 512
+    cfg.model.trunk_sizes = (512, 512)
+    cfg.model.dtype = "bf16"
+    cfg.replay.seq_len = 80
+    cfg.replay.burn_in = 40
+    cfg.learner.batch_size = 256
+    spec = make_env(cfg.env.name).spec
+    torch.manual_seed(0)
+    lrn = Learner(cfg, spec, device="cuda")
+    lrn.collect_stats = False
+    B, L, H = 256, 85, 512
+    g = torch.Generator().manual_seed(0)
+    batch = {
+        "obs": torch.randn(B, L, *spec.obs_shape, generator=g).cuda(),
+        "action": torch.randn(B, L, spec.action_dim, generator=g).clamp_(-1, 1).cuda(),
+        "reward": torch.rand(B, L, generator=g).cuda(),
+        "done": torch.zeros(B, L).cuda(),
+        "state": torch.zeros(B, 4, H).cuda(),
+        "weight": torch.ones(B).cuda(),
+        "idx": None,
+    }
+    for _ in range(8):
+        lrn.train_step(batch)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    n = 25
+    for _ in range(n):
+        lrn.train_step(batch)
+    torch.cuda.synchronize()
+    tps = n * B * cfg.train_len / (time.perf_counter() - t0)
+    assert ext().lstm_px_error() == 0
+    assert tps > 1.2e6, f"flagship step at {tps:.0f} transitions/s — " \
+        "far below the native-path floor (silent fallback or livelock?)"

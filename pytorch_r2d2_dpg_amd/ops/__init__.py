@@ -1,12 +1,13 @@
 """Hot-op layer.
 
-Every op has exactly two implementations and NO runtime multi-backend
-dispatch (north-star constraint):
-
-* a plain PyTorch fp32 reference in :mod:`.reference` — the CPU path and the
-  numerics oracle for tests,
-* a hand-written CDNA4 HIP kernel in ``csrc/`` (gfx950-only), loaded from the
-  in-tree extension ``_r2d2dpg_hip``.
+Every op has a plain PyTorch fp32 reference in :mod:`.reference` — the CPU
+path and the numerics oracle for tests — and hand-written CDNA4 HIP kernels
+in ``csrc/`` (gfx950-only), loaded from the in-tree extension
+``_r2d2dpg_hip``. There is NO runtime multi-backend dispatch (north-star
+constraint): on GPU the HIP path is the only path. Where csrc keeps an
+alternate NATIVE kernel for an op (slab vs granule backward exchange,
+pipelined forward), selection is a process-wide env choice measured and
+documented in docs/KERNELS.md, never a per-call fallback.
 
 On a GPU device the HIP extension is REQUIRED: if a tensor is on a HIP device
 and the extension is missing, ops raise instead of silently falling back to

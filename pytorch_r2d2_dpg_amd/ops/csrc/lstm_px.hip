@@ -236,7 +236,12 @@ lstm_fwd_px_kernel(const __hip_bfloat16* __restrict__ gates_x,  // [T,B,4H]
                 }
             }
         }
-        __syncthreads();  // next sweep writes the other LDS buffer
+        // NO trailing barrier: the next sweep writes the OTHER LDS buffer,
+        // and a wave cannot get two steps ahead — its next-next sweep would
+        // need granules its own block's slow waves publish only after they
+        // pass the (kept) pre-MFMA barrier. The double buffer + tag polls
+        // bound the skew to one buffer; removing this barrier saves one
+        // block-wide sync per timestep.
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -1158,7 +1163,7 @@ lstm_fwd_px2_kernel(PxArgs a0, PxArgs a1, unsigned int* err, int B, int H) {
                 }
             }
         }
-        __syncthreads();
+        // no trailing barrier — see the solo forward kernel
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
